@@ -1,0 +1,85 @@
+"""Distributed data-parallel logic on CPU (gloo, world_size=2).
+
+Verifies the bucketer's all-reduce correctness: two ranks with identical
+replicas and different minibatches must end up with identical summed
+grads, equal to the single-process sum of the two per-batch grads — the
+1-GPU vs N-GPU gradient-equality contract (SURVEY.md §4 'Distributed').
+"""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as td
+import torch.multiprocessing as mp
+
+from zaremba_amd import trainer
+from zaremba_amd.models.lstm_lm import Model
+from zaremba_amd.ops import functional as F_ref
+
+V, H, L, B, T = 23, 8, 2, 3, 4
+
+
+def _make_batch(seed):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randint(0, V, (T, B), generator=g)
+    y = torch.randint(0, V, (T, B), generator=g)
+    return x, y
+
+
+def _rank_main(rank, world, init_file, q):
+    td.init_process_group("gloo", init_method=f"file://{init_file}",
+                          rank=rank, world_size=world)
+    torch.manual_seed(7)
+    model = Model(V, H, L, dropout=0.0, winit=0.1)
+    from zaremba_amd.parallel.bucketer import GradBucketer
+    dp = GradBucketer(model, bucket_bytes=4096)  # force several buckets
+    x, y = _make_batch(100 + rank)
+    states = model.state_init(B)
+    dp.zero_grad()
+    scores, _ = model(x, states)
+    loss = F_ref.nll_loss(scores, y)
+    loss.backward()
+    dp.finalize_backward()
+    # ship by value (numpy) — CUDA/file-descriptor tensor sharing is not
+    # reliable across short-lived spawn workers
+    grads = {n: p.grad.numpy().copy() for n, p in model.named_parameters()}
+    q.put((rank, grads))
+    td.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_bucketer_allreduce_matches_serial_sum():
+    with tempfile.TemporaryDirectory() as d:
+        init_file = os.path.join(d, "pg")
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        procs = [ctx.Process(target=_rank_main, args=(r, 2, init_file, q))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(2):
+            rank, grads = q.get()
+            results[rank] = grads
+        for p in procs:
+            p.join(60)
+            assert p.exitcode == 0
+
+    # serial reference: same init, grads of batch0 + grads of batch1
+    torch.manual_seed(7)
+    model = Model(V, H, L, dropout=0.0, winit=0.1)
+    expected = {n: torch.zeros_like(p) for n, p in model.named_parameters()}
+    for seed in (100, 101):
+        model.zero_grad()
+        x, y = _make_batch(seed)
+        scores, _ = model(x, model.state_init(B))
+        F_ref.nll_loss(scores, y).backward()
+        for n, p in model.named_parameters():
+            expected[n] += p.grad
+    for n in expected:
+        r0 = torch.from_numpy(results[0][n])
+        r1 = torch.from_numpy(results[1][n])
+        assert torch.allclose(r0, expected[n], atol=1e-5), n
+        assert torch.allclose(r0, r1, atol=1e-7), n

@@ -1,0 +1,95 @@
+"""Bucketed gradient all-reduce overlapped with the BPTT backward.
+
+Design (MI355X/xGMI-first, SURVEY.md §5 'Distributed communication'):
+  * parameters are grouped into buckets in REVERSE parameter order — the
+    order backward produces grads (fc -> layer L..1 -> embed) — so each
+    bucket's all-reduce launches while earlier layers are still running
+    their backward kernels,
+  * each bucket owns one persistent flat fp32 buffer; ``param.grad`` is a
+    view into it, so autograd accumulates straight into the communication
+    buffer (no pack/unpack copies),
+  * the per-param post-accumulate-grad hook counts the bucket down and
+    fires an async RCCL all-reduce (SUM) the moment the bucket completes;
+    ``finalize_backward()`` joins all in-flight works before the fused
+    clip+SGD consumes the grads (which divides by world_size),
+  * xGMI is point-to-point (7 links/GPU): grads for the Large model are
+    only ~266 MB fp32, so latency and overlap dominate — the default
+    bucket size (25 MB) keeps a handful of in-flight collectives without
+    fragmenting into launch-bound slivers.
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.distributed as td
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter]):
+        self.params = params
+        total = sum(p.numel() for p in params)
+        dev = params[0].device
+        self.flat = torch.zeros(total, dtype=torch.float32, device=dev)
+        offset = 0
+        for p in params:
+            n = p.numel()
+            p.grad = self.flat[offset:offset + n].view_as(p)
+            offset += n
+        self.pending = len(params)
+        self.work = None
+
+    def reset(self):
+        self.pending = len(self.params)
+        self.work = None
+
+
+class GradBucketer:
+    def __init__(self, model: torch.nn.Module, bucket_bytes: int = 25 << 20):
+        if not (td.is_available() and td.is_initialized()):
+            raise RuntimeError("GradBucketer requires an initialized process group")
+        self.world_size = td.get_world_size()
+        params = [p for p in model.parameters() if p.requires_grad]
+        params.reverse()  # backward completion order
+        self.buckets: List[_Bucket] = []
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        for p in params:
+            cur.append(p)
+            cur_bytes += p.numel() * 4
+            if cur_bytes >= bucket_bytes:
+                self.buckets.append(_Bucket(cur))
+                cur, cur_bytes = [], 0
+        if cur:
+            self.buckets.append(_Bucket(cur))
+        self._by_param = {}
+        self._hooks = []
+        for b in self.buckets:
+            for p in b.params:
+                self._by_param[p] = b
+                self._hooks.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    def _on_grad_ready(self, param: torch.nn.Parameter):
+        b = self._by_param[param]
+        b.pending -= 1
+        if b.pending == 0:
+            b.work = td.all_reduce(b.flat, op=td.ReduceOp.SUM, async_op=True)
+
+    def zero_grad(self):
+        for b in self.buckets:
+            b.reset()
+
+    def finalize_backward(self):
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+            elif b.pending != 0:
+                # A parameter produced no grad this step (should not happen in
+                # this model); reduce anyway so replicas stay in sync.
+                td.all_reduce(b.flat, op=td.ReduceOp.SUM)
+
+    def detach_hooks(self):
+        for h in self._hooks:
+            h.remove()
