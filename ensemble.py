@@ -110,6 +110,8 @@ def train_one(args, vocab_size, data, model_num):
 
 def main():
     args = build_parser().parse_args()
+    if args.engine == "eager":
+        os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
     setdevice(args, plural=True)
     if zdist.is_rank0():

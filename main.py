@@ -109,6 +109,8 @@ def load_data(args):
 
 def main():
     args = build_parser().parse_args()
+    if args.engine == "eager":
+        os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
     if args.seed is not None:
         torch.manual_seed(args.seed + zdist.rank())

@@ -1,0 +1,295 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+of the same op (SURVEY.md §4 'Kernel unit tests'). Run on MI355X via
+`pytest -m gpu`.
+
+Tolerances: bf16 inputs / fp32 accumulation — elementwise ops ~1e-2
+relative; GEMMs scale with sqrt(K) * bf16 eps.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from zaremba_amd import _C
+    return _C.ext()
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def rel_err(a, b):
+    a = a.float()
+    b = b.float()
+    return ((a - b).abs() / (b.abs().clamp_min(1.0))).max().item()
+
+
+# ---------------------------------------------------------------------------
+# GEMM
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (700, 6000, 1500),
+                                   (700, 10000, 1500), (37, 129, 65)])
+def test_gemm_nt(ext, M, N, K):
+    torch.manual_seed(0)
+    A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
+    B = torch.randn(N, K, device=dev(), dtype=torch.bfloat16)
+    bias = torch.randn(N, device=dev(), dtype=torch.float32)
+    C = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    ext.gemm(A, B, C, bias, False, False)
+    ref = A.float() @ B.float().t() + bias
+    tol = 3e-2 * math.sqrt(K) / 10
+    assert rel_err(C, ref) < max(tol, 3e-2), rel_err(C, ref)
+
+
+def test_gemm_nt_asymmetric_detects_transpose(ext):
+    """A = one-hot-ish, asymmetric B: catches swapped C row/col mapping."""
+    M = N = K = 64
+    A = torch.zeros(M, K, device=dev(), dtype=torch.bfloat16)
+    for i in range(M):
+        A[i, i % K] = 1.0
+    B = torch.arange(N * K, device=dev(), dtype=torch.float32).reshape(N, K)
+    B = (B % 37).to(torch.bfloat16)
+    C = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    ext.gemm(A, B, C, None, False, False)
+    ref = A.float() @ B.float().t()
+    assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (6000, 1500, 700),
+                                   (10000, 1500, 700), (100, 61, 35)])
+def test_gemm_tn(ext, M, N, K):
+    torch.manual_seed(1)
+    A = torch.randn(K, M, device=dev(), dtype=torch.bfloat16)
+    B = torch.randn(K, N, device=dev(), dtype=torch.bfloat16)
+    C = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    ext.gemm(A, B, C, None, True, True)
+    ref = A.float().t() @ B.float()
+    assert rel_err(C, ref) < 3e-2, rel_err(C, ref)
+
+
+def test_gemm_bf16_out(ext):
+    torch.manual_seed(2)
+    M, N, K = 130, 140, 96
+    A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
+    B = torch.randn(N, K, device=dev(), dtype=torch.bfloat16)
+    C = torch.empty(M, N, device=dev(), dtype=torch.bfloat16)
+    ext.gemm(A, B, C, None, False, False)
+    ref = A.float() @ B.float().t()
+    assert rel_err(C, ref) < 5e-2
+
+
+def test_smallm_gemm_nt(ext):
+    torch.manual_seed(3)
+    M, N, K = 20, 1500, 6000
+    A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
+    B = torch.randn(N, K, device=dev(), dtype=torch.bfloat16)
+    C = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    ext.smallm_gemm_nt(A, B, C)
+    ref = A.float() @ B.float().t()
+    assert rel_err(C, ref) < 5e-2, rel_err(C, ref)
+
+
+# ---------------------------------------------------------------------------
+# LSTM cell + sequence
+# ---------------------------------------------------------------------------
+def _ref_cell(h, c, gx, W_h):
+    """fp32 oracle of the fused cell (reference model.py:34-45 with the
+    input-side gates + both biases pre-folded into gx)."""
+    g = gx.float() + h.float() @ W_h.float().t()
+    H = h.size(1)
+    gi, gf, go, gn = g.split(H, dim=1)
+    i = torch.sigmoid(gi)
+    f = torch.sigmoid(gf)
+    o = torch.sigmoid(go)
+    n = torch.tanh(gn)
+    c2 = f * c.float() + i * n
+    h2 = o * torch.tanh(c2)
+    return h2, c2, (i, f, o, n)
+
+
+@pytest.mark.parametrize("B,H", [(20, 1500), (20, 650), (7, 200), (20, 100)])
+def test_lstm_cell_fwd_step(ext, B, H):
+    torch.manual_seed(4)
+    h = (torch.randn(B, H, device=dev()) * 0.5).to(torch.bfloat16)
+    c = torch.randn(B, H, device=dev()) * 0.5
+    gx = (torch.randn(B, 4 * H, device=dev()) * 0.5).to(torch.bfloat16)
+    W_h = (torch.randn(4 * H, H, device=dev()) * 0.02).to(torch.bfloat16)
+    h_out = torch.empty(B, H, device=dev(), dtype=torch.bfloat16)
+    c_out = torch.empty(B, H, device=dev(), dtype=torch.float32)
+    gates = torch.empty(B, 4 * H, device=dev(), dtype=torch.bfloat16)
+    ext.lstm_cell_fwd_step(h, c, gx, W_h, h_out, c_out, gates)
+    h_ref, c_ref, (i, f, o, n) = _ref_cell(h, c, gx, W_h)
+    assert rel_err(c_out, c_ref) < 2e-2, rel_err(c_out, c_ref)
+    assert rel_err(h_out, h_ref) < 2e-2
+    gr = torch.cat([i, f, o, n], dim=1)
+    assert rel_err(gates, gr) < 2e-2
+
+
+def test_lstm_seq_fwd_matches_step_loop(ext):
+    torch.manual_seed(5)
+    T, B, H = 6, 20, 650
+    gx = (torch.randn(T, B, 4 * H, device=dev()) * 0.5).to(torch.bfloat16)
+    W_h = (torch.randn(4 * H, H, device=dev()) * 0.02).to(torch.bfloat16)
+    h_all = torch.zeros(T + 1, B, H, device=dev(), dtype=torch.bfloat16)
+    c_all = torch.zeros(T + 1, B, H, device=dev(), dtype=torch.float32)
+    gates = torch.empty(T, B, 4 * H, device=dev(), dtype=torch.bfloat16)
+    h_all[0] = (torch.randn(B, H, device=dev()) * 0.3).to(torch.bfloat16)
+    c_all[0] = torch.randn(B, H, device=dev()) * 0.3
+    ext.lstm_seq_fwd(gx, W_h, h_all, c_all, gates)
+    h = h_all[0]
+    c = c_all[0].clone()
+    for t in range(T):
+        h_ref, c_ref, _ = _ref_cell(h, c, gx[t], W_h)
+        assert rel_err(h_all[t + 1], h_ref) < 3e-2, t
+        h = h_all[t + 1]  # carry the kernel's bf16 h to isolate per-step err
+        c = c_all[t + 1].clone()
+
+
+def test_lstm_layer_autograd_matches_eager(ext):
+    """Full layer fwd+bwd through LstmLayerFn vs fp32 autograd oracle."""
+    from zaremba_amd.models.lstm_lm import Model
+    torch.manual_seed(6)
+    T, B, H, V = 8, 20, 200, 50
+    model = Model(V, H, 1, dropout=0.0, winit=0.05, lstm_type="custom",
+                  engine="hip").to(dev())
+    ref = Model(V, H, 1, dropout=0.0, winit=0.05, lstm_type="custom",
+                engine="eager").to(dev())
+    ref.load_state_dict(model.state_dict())
+    x = torch.randint(0, V, (T, B), device=dev())
+    y = torch.randint(0, V, (T, B), device=dev())
+
+    from zaremba_amd import trainer
+    model.train(), ref.train()
+    s1 = model.state_init(B)
+    scores1, s1 = model(x, s1)
+    loss1 = trainer.nll_loss(scores1, y)
+    loss1.backward()
+    import os
+    os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
+    try:
+        s2 = ref.state_init(B)
+        scores2, s2 = ref(x, s2)
+        loss2 = trainer.nll_loss(scores2, y)
+        loss2.backward()
+    finally:
+        del os.environ["ZAREMBA_AMD_FORCE_EAGER"]
+
+    assert rel_err(scores1, scores2) < 5e-2, rel_err(scores1, scores2)
+    assert abs(loss1.item() - loss2.item()) / abs(loss2.item()) < 2e-2
+    for (n1, p1), (n2, p2) in zip(model.named_parameters(),
+                                  ref.named_parameters()):
+        e = rel_err(p1.grad, p2.grad)
+        assert e < 8e-2, (n1, e)
+
+
+# ---------------------------------------------------------------------------
+# Embedding
+# ---------------------------------------------------------------------------
+def test_embedding_fwd_bwd(ext):
+    torch.manual_seed(7)
+    V, H, N = 1000, 1500, 700
+    W = torch.randn(V, H, device=dev(), dtype=torch.bfloat16)
+    idx = torch.randint(0, V, (N,), device=dev())
+    out = torch.empty(N, H, device=dev(), dtype=torch.bfloat16)
+    ext.embedding_fwd(W, idx, out)
+    assert torch.equal(out, W[idx])
+    dY = torch.randn(N, H, device=dev(), dtype=torch.bfloat16)
+    dW = torch.zeros(V, H, device=dev(), dtype=torch.float32)
+    ext.embedding_bwd(dY, idx, dW)
+    ref = torch.zeros(V, H, device=dev(), dtype=torch.float32)
+    ref.index_add_(0, idx, dY.float())
+    assert rel_err(dW, ref) < 1e-2
+
+
+# ---------------------------------------------------------------------------
+# Dropout
+# ---------------------------------------------------------------------------
+def test_dropout_stats_and_mask_replay(ext):
+    torch.manual_seed(8)
+    n = 1_000_000
+    p = 0.35
+    x = torch.ones(n, device=dev(), dtype=torch.bfloat16)
+    y = torch.empty_like(x)
+    counter = torch.zeros(1, dtype=torch.int64, device=dev())
+    off = torch.zeros(1, dtype=torch.int64, device=dev())
+    ext.dropout_fwd(x, y, p, 1234, counter, off)
+    yk = y.float()
+    keep_frac = (yk != 0).float().mean().item()
+    assert abs(keep_frac - (1 - p)) < 5e-3
+    scale = yk[yk != 0].mean().item()
+    assert abs(scale - 1.0 / (1 - p)) < 1e-2
+    # backward regenerates the same mask
+    dy = torch.ones_like(x)
+    dx = torch.empty_like(x)
+    ext.dropout_bwd(dy, dx, p, 1234, off)
+    assert torch.equal((dx != 0), (y != 0))
+    # counter advanced; a second fwd draws a different mask
+    y2 = torch.empty_like(x)
+    ext.dropout_fwd(x, y2, p, 1234, counter, off)
+    assert not torch.equal((y2 != 0), (y != 0))
+
+
+# ---------------------------------------------------------------------------
+# Loss
+# ---------------------------------------------------------------------------
+def test_lsm_nll_fwd_bwd_vs_eager(ext):
+    torch.manual_seed(9)
+    N, V, B = 700, 10000, 20
+    scores = (torch.randn(N, V, device=dev()) * 3).requires_grad_(True)
+    y = torch.randint(0, V, (N // B, B), device=dev())
+    from zaremba_amd.ops.hip_ops import nll_loss_hip
+    from zaremba_amd.ops import functional as F_ref
+    loss = nll_loss_hip(scores, y, B)
+    loss.backward()
+    g1 = scores.grad.clone()
+    scores2 = scores.detach().clone().requires_grad_(True)
+    loss2 = F_ref.nll_loss(scores2, y)
+    loss2.backward()
+    assert abs(loss.item() - loss2.item()) / loss2.item() < 1e-4
+    assert (g1 - scores2.grad).abs().max().item() < 1e-5
+
+
+# ---------------------------------------------------------------------------
+# SGD
+# ---------------------------------------------------------------------------
+def test_fused_clip_sgd_matches_eager(ext):
+    torch.manual_seed(10)
+    sizes = [(100, 64), (256,), (64, 100)]
+    masters = [torch.randn(*s, device=dev()) for s in sizes]
+    grads = [torch.randn(*s, device=dev()) * 5 for s in sizes]
+    # eager reference
+    import torch.nn as nn
+    ps = [nn.Parameter(m.clone()) for m in masters]
+    for p, g in zip(ps, grads):
+        p.grad = g.clone()
+    ref_norm = nn.utils.clip_grad_norm_(ps, 2.0)
+    with torch.no_grad():
+        for p in ps:
+            p -= 0.3 * p.grad
+    # fused kernels
+    norm2 = torch.zeros(1, device=dev())
+    for g in grads:
+        ext.norm2_accum(g.reshape(-1), norm2)
+    assert abs(norm2.sqrt().item() - ref_norm.item()) / ref_norm.item() < 1e-5
+    shadows = [torch.empty(m.numel(), device=dev(), dtype=torch.bfloat16)
+               for m in masters]
+    for m, g, sh in zip(masters, grads, shadows):
+        ext.sgd_update(m.view(-1), g.reshape(-1), sh, norm2, 2.0, 0.3, 1.0)
+    for m, p, sh in zip(masters, ps, shadows):
+        assert torch.allclose(m, p.detach(), atol=1e-5)
+        assert rel_err(sh.view(m.shape), m) < 1e-2
+
+
+def test_transpose_bf16(ext):
+    torch.manual_seed(11)
+    for R, C in [(6000, 1500), (127, 33), (64, 64)]:
+        src = torch.randn(R, C, device=dev(), dtype=torch.bfloat16)
+        dst = torch.empty(C, R, device=dev(), dtype=torch.bfloat16)
+        ext.transpose_bf16(src, dst)
+        assert torch.equal(dst, src.t().contiguous())

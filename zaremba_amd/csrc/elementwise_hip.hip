@@ -1,0 +1,298 @@
+#include "hip/hip_runtime.h"
+// Elementwise / reduction kernels for gfx950:
+//   K1  embedding gather fwd + fp32 scatter-add bwd (reference model.py:14)
+//   K5  philox inverted dropout, mask regenerated in bwd (model.py:87,105,108)
+//   K7  fused log-softmax + NLL fwd/bwd (reference main.py:77-84, stable form)
+//   K9/K10 fused grad-norm^2 reduce + clipped SGD with fp32 master weights
+//          and bf16 shadow rewrite (reference main.py:115-117)
+#include "common.h"
+#include "philox.h"
+
+namespace zamd {
+
+// ---------------------------------------------------------------------------
+// K1: embedding
+// ---------------------------------------------------------------------------
+__global__ void embedding_fwd_kernel(const bf16* __restrict__ W,
+                                     const int64_t* __restrict__ idx,
+                                     bf16* __restrict__ out, int N, int H) {
+  // one wave per token row; 16B vector copies
+  int row = blockIdx.x * (blockDim.x / 64) + wave_id();
+  if (row >= N) return;
+  const bf16* src = W + (int64_t)idx[row] * H;
+  bf16* dst = out + (int64_t)row * H;
+  int l = lane_id();
+  int nv = H / 8;
+  for (int v = l; v < nv; v += 64)
+    reinterpret_cast<bf16x8*>(dst)[v] =
+        reinterpret_cast<const bf16x8*>(src)[v];
+  for (int k = nv * 8 + l; k < H; k += 64) dst[k] = src[k];
+}
+
+void launch_embedding_fwd(const bf16* W, const int64_t* idx, bf16* out,
+                          int N, int H, hipStream_t stream) {
+  int waves_per_block = 4;
+  int grid = cdiv(N, waves_per_block);
+  hipLaunchKernelGGL(embedding_fwd_kernel, dim3(grid),
+                     dim3(waves_per_block * 64), 0, stream, W, idx, out, N, H);
+}
+
+__global__ void embedding_bwd_kernel(const bf16* __restrict__ dY,
+                                     const int64_t* __restrict__ idx,
+                                     float* __restrict__ dW, int N, int H) {
+  int row = blockIdx.x * (blockDim.x / 64) + wave_id();
+  if (row >= N) return;
+  const bf16* src = dY + (int64_t)row * H;
+  float* dst = dW + (int64_t)idx[row] * H;
+  for (int k = lane_id(); k < H; k += 64)
+    atomicAdd(dst + k, bf2f(src[k]));
+}
+
+void launch_embedding_bwd(const bf16* dY, const int64_t* idx, float* dW,
+                          int N, int H, hipStream_t stream) {
+  int waves_per_block = 4;
+  int grid = cdiv(N, waves_per_block);
+  hipLaunchKernelGGL(embedding_bwd_kernel, dim3(grid),
+                     dim3(waves_per_block * 64), 0, stream, dY, idx, dW, N, H);
+}
+
+// ---------------------------------------------------------------------------
+// K5: dropout (inverted, scale 1/(1-p) at train time)
+// ---------------------------------------------------------------------------
+// The fwd kernel draws its philox offset from a device counter (so graph
+// replays advance the stream), saves the offset it used for backward, and
+// bumps the counter. 4 elements per philox call.
+__global__ void dropout_fwd_kernel(const bf16* __restrict__ x,
+                                   bf16* __restrict__ y, float p,
+                                   uint64_t seed,
+                                   uint64_t* __restrict__ counter,
+                                   uint64_t* __restrict__ saved_offset,
+                                   int64_t n) {
+  __shared__ uint64_t off_s;
+  if (threadIdx.x == 0) off_s = *saved_offset;
+  __syncthreads();
+  uint64_t off = off_s;
+  float scale = 1.f / (1.f - p);
+  int64_t quad = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nquads = (n + 3) / 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; quad < nquads; quad += stride) {
+    Philox4 r = philox4x32_10(seed, off + quad);
+    uint32_t u[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int64_t i = quad * 4 + e;
+      if (i < n) {
+        bool keep = u32_to_uniform(u[e]) >= p;
+        y[i] = keep ? f2bf(bf2f(x[i]) * scale) : (bf16)0.f;
+      }
+    }
+  }
+}
+
+// Host-side step: latch the counter into saved_offset, then advance it.
+__global__ void dropout_tick_kernel(uint64_t* counter, uint64_t* saved_offset,
+                                    int64_t nquads) {
+  *saved_offset = *counter;
+  *counter += nquads;
+}
+
+__global__ void dropout_bwd_kernel(const bf16* __restrict__ dy,
+                                   bf16* __restrict__ dx, float p,
+                                   uint64_t seed,
+                                   const uint64_t* __restrict__ saved_offset,
+                                   int64_t n) {
+  __shared__ uint64_t off_s;
+  if (threadIdx.x == 0) off_s = *saved_offset;
+  __syncthreads();
+  uint64_t off = off_s;
+  float scale = 1.f / (1.f - p);
+  int64_t quad = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nquads = (n + 3) / 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; quad < nquads; quad += stride) {
+    Philox4 r = philox4x32_10(seed, off + quad);
+    uint32_t u[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int64_t i = quad * 4 + e;
+      if (i < n) {
+        bool keep = u32_to_uniform(u[e]) >= p;
+        dx[i] = keep ? f2bf(bf2f(dy[i]) * scale) : (bf16)0.f;
+      }
+    }
+  }
+}
+
+static int dropout_grid(int64_t n) {
+  int64_t nquads = (n + 3) / 4;
+  int g = cdiv(nquads, 256);
+  return g > 2048 ? 2048 : g;
+}
+
+void launch_dropout_fwd(const bf16* x, bf16* y, float p, uint64_t seed,
+                        uint64_t* counter, uint64_t* saved_offset, int64_t n,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(dropout_tick_kernel, dim3(1), dim3(1), 0, stream,
+                     counter, saved_offset, (n + 3) / 4);
+  hipLaunchKernelGGL(dropout_fwd_kernel, dim3(dropout_grid(n)), dim3(256), 0,
+                     stream, x, y, p, seed, counter, saved_offset, n);
+}
+
+void launch_dropout_bwd(const bf16* dy, bf16* dx, float p, uint64_t seed,
+                        const uint64_t* saved_offset, int64_t n,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(dropout_bwd_kernel, dim3(dropout_grid(n)), dim3(256), 0,
+                     stream, dy, dx, p, seed, saved_offset, n);
+}
+
+// ---------------------------------------------------------------------------
+// K7: fused log-softmax + NLL (stable; scores fp32 [N, V])
+// ---------------------------------------------------------------------------
+// fwd: per row, lse = max + log(sum exp(s - max)); atomically accumulates
+// sum_n (lse_n - s_n[y_n]) into loss_accum (zeroed by the host wrapper);
+// saves lse for backward. Loss scale (batch_size / N) applied host-side
+// as a lazy torch op.
+__global__ void lsm_nll_fwd_kernel(const float* __restrict__ scores,
+                                   const int64_t* __restrict__ y,
+                                   float* __restrict__ lse,
+                                   float* __restrict__ loss_accum, int N,
+                                   int V) {
+  __shared__ float scratch[8];
+  int row = blockIdx.x;
+  if (row >= N) return;
+  const float* s = scores + (int64_t)row * V;
+  float m = -INFINITY;
+  for (int v = threadIdx.x; v < V; v += blockDim.x) m = fmaxf(m, s[v]);
+  m = block_reduce(m, scratch, -INFINITY,
+                   [] __device__(float a, float b) { return fmaxf(a, b); });
+  __syncthreads();
+  if (threadIdx.x == 0) scratch[0] = m;
+  __syncthreads();
+  m = scratch[0];
+  float acc = 0.f;
+  for (int v = threadIdx.x; v < V; v += blockDim.x) acc += __expf(s[v] - m);
+  __syncthreads();
+  acc = block_reduce(acc, scratch, 0.f,
+                     [] __device__(float a, float b) { return a + b; });
+  if (threadIdx.x == 0) {
+    float l = m + __logf(acc);
+    lse[row] = l;
+    atomicAdd(loss_accum, l - s[y[row]]);
+  }
+}
+
+void launch_lsm_nll_fwd(const float* scores, const int64_t* y, float* lse,
+                        float* loss_accum, int N, int V, hipStream_t stream) {
+  hipLaunchKernelGGL(lsm_nll_fwd_kernel, dim3(N), dim3(256), 0, stream,
+                     scores, y, lse, loss_accum, N, V);
+}
+
+// bwd: dscores = (softmax - onehot_y) * gscale, gscale = upstream * B / N
+// (upstream read from a device scalar so no sync is needed).
+__global__ void lsm_nll_bwd_kernel(const float* __restrict__ scores,
+                                   const float* __restrict__ lse,
+                                   const int64_t* __restrict__ y,
+                                   const float* __restrict__ upstream,
+                                   float scale, float* __restrict__ dscores,
+                                   int N, int V) {
+  int row = blockIdx.x;
+  if (row >= N) return;
+  float g = *upstream * scale;
+  const float* s = scores + (int64_t)row * V;
+  float* d = dscores + (int64_t)row * V;
+  float l = lse[row];
+  int64_t yy = y[row];
+  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+    float p = __expf(s[v] - l);
+    d[v] = (p - (v == yy ? 1.f : 0.f)) * g;
+  }
+}
+
+void launch_lsm_nll_bwd(const float* scores, const float* lse,
+                        const int64_t* y, const float* upstream, float scale,
+                        float* dscores, int N, int V, hipStream_t stream) {
+  hipLaunchKernelGGL(lsm_nll_bwd_kernel, dim3(N), dim3(256), 0, stream,
+                     scores, lse, y, upstream, scale, dscores, N, V);
+}
+
+// ---------------------------------------------------------------------------
+// K9/K10: fused grad clip + SGD
+// ---------------------------------------------------------------------------
+__global__ void norm2_accum_kernel(const float* __restrict__ g, int64_t n,
+                                   float* __restrict__ accum) {
+  __shared__ float scratch[8];
+  float acc = 0.f;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float v = g[i];
+    acc += v * v;
+  }
+  acc = block_reduce(acc, scratch, 0.f,
+                     [] __device__(float a, float b) { return a + b; });
+  if (threadIdx.x == 0) atomicAdd(accum, acc);
+}
+
+void launch_norm2_accum(const float* g, int64_t n, float* accum,
+                        hipStream_t stream) {
+  int grid = cdiv(n, 256);
+  if (grid > 2048) grid = 2048;
+  hipLaunchKernelGGL(norm2_accum_kernel, dim3(grid), dim3(256), 0, stream, g,
+                     n, accum);
+}
+
+// master -= lr * grad_scale * clip(norm) * grad; shadow/bf16 rewritten in
+// the same pass. `shadow` may be null (biases stay fp32).
+__global__ void sgd_update_kernel(float* __restrict__ master,
+                                  const float* __restrict__ grad,
+                                  bf16* __restrict__ shadow,
+                                  const float* __restrict__ norm2,
+                                  float max_norm, float lr, float grad_scale,
+                                  int64_t n) {
+  float norm = sqrtf(*norm2) * grad_scale;
+  float coef = max_norm / (norm + 1e-6f);
+  coef = fminf(coef, 1.f) * lr * grad_scale;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float v = master[i] - coef * grad[i];
+    master[i] = v;
+    if (shadow) shadow[i] = f2bf(v);
+  }
+}
+
+void launch_sgd_update(float* master, const float* grad, bf16* shadow,
+                       const float* norm2, float max_norm, float lr,
+                       float grad_scale, int64_t n, hipStream_t stream) {
+  int grid = cdiv(n, 256);
+  if (grid > 2048) grid = 2048;
+  hipLaunchKernelGGL(sgd_update_kernel, dim3(grid), dim3(256), 0, stream,
+                     master, grad, shadow, norm2, max_norm, lr, grad_scale, n);
+}
+
+// transposed bf16 shadow refresh: dst[c][r] = src[r][c] (LDS-tiled)
+__global__ void transpose_bf16_kernel(const bf16* __restrict__ src,
+                                      bf16* __restrict__ dst, int R, int C) {
+  __shared__ bf16 tile[32][33];
+  int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
+  int tc = threadIdx.x % 32, tr = threadIdx.x / 32;  // 32x8 threads
+  for (int i = 0; i < 32; i += 8) {
+    int r = r0 + tr + i, c = c0 + tc;
+    tile[tr + i][tc] = (r < R && c < C) ? src[(int64_t)r * C + c] : (bf16)0.f;
+  }
+  __syncthreads();
+  for (int i = 0; i < 32; i += 8) {
+    int r = r0 + tc, c = c0 + tr + i;  // write transposed, coalesced in r
+    if (r < R && c < C) dst[(int64_t)c * R + r] = tile[tc][tr + i];
+  }
+}
+
+void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
+                           hipStream_t stream) {
+  dim3 grid(cdiv(C, 32), cdiv(R, 32));
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream, src,
+                     dst, R, C);
+}
+
+}  // namespace zamd

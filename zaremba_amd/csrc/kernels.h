@@ -1,0 +1,55 @@
+// Host-side launcher prototypes for the gfx950 kernel library.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace zamd {
+
+using bf16 = __bf16;
+
+// gemm.hip — C[M,N] = A' @ B' (+bias); TRANS_A: A is [K,M]; TRANS_B: B is
+// [K,N] (otherwise the NT weight layout B[N,K]).
+template <bool TA, bool TB, typename OutT>
+void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
+                   int M, int N, int K, int lda, int ldb, int ldc,
+                   hipStream_t stream);
+
+// lstm.hip
+void launch_lstm_cell_fwd(const bf16* h_prev, const float* c_prev,
+                          const bf16* gx, const bf16* W_h, bf16* h_out,
+                          float* c_out, bf16* gates_out, int B, int H,
+                          hipStream_t stream);
+void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
+                              const bf16* gates, const float* c_prev,
+                              const float* c_new, bf16* dG, int B, int H,
+                              hipStream_t stream);
+void launch_smallm_gemm_nt(const bf16* A, const bf16* B, float* C, int M,
+                           int N, int K, hipStream_t stream);
+
+// elementwise.hip
+void launch_embedding_fwd(const bf16* W, const int64_t* idx, bf16* out,
+                          int N, int H, hipStream_t stream);
+void launch_embedding_bwd(const bf16* dY, const int64_t* idx, float* dW,
+                          int N, int H, hipStream_t stream);
+void launch_dropout_fwd(const bf16* x, bf16* y, float p, uint64_t seed,
+                        uint64_t* counter, uint64_t* saved_offset, int64_t n,
+                        hipStream_t stream);
+void launch_dropout_bwd(const bf16* dy, bf16* dx, float p, uint64_t seed,
+                        const uint64_t* saved_offset, int64_t n,
+                        hipStream_t stream);
+void launch_lsm_nll_fwd(const float* scores, const int64_t* y, float* lse,
+                        float* loss_accum, int N, int V, hipStream_t stream);
+void launch_lsm_nll_bwd(const float* scores, const float* lse,
+                        const int64_t* y, const float* upstream, float scale,
+                        float* dscores, int N, int V, hipStream_t stream);
+void launch_norm2_accum(const float* g, int64_t n, float* accum,
+                        hipStream_t stream);
+void launch_sgd_update(float* master, const float* grad, bf16* shadow,
+                       const float* norm2, float max_norm, float lr,
+                       float grad_scale, int64_t n, hipStream_t stream);
+void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
+                           hipStream_t stream);
+
+}  // namespace zamd

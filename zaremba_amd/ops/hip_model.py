@@ -1,0 +1,181 @@
+"""HipModel — the MI355X execution runtime attached to a Model.
+
+Owns, per model instance:
+  * bf16 shadow weights (+ pre-transposed shadows for the backward data
+    GEMMs, and the folded b_x+b_h bias vector the input GEMM consumes),
+    kept in sync with the fp32 masters by the fused clip+SGD kernels,
+  * persistent per-layer workspaces (gx, h_all, c_all, gates, dG, ...)
+    sized on first use so the C++ LSTM sequence drivers can hipGraph-
+    capture the per-timestep launch trains against stable pointers,
+  * the philox dropout counter (device-side, graph-replay safe),
+  * the fused clip+SGD step (grad-norm^2 reduce -> per-param update
+    rewriting master fp32 + bf16 shadow in one pass -> transposed-shadow
+    refresh kernels).
+"""
+
+from __future__ import annotations
+
+import secrets
+from typing import List, Optional
+
+import torch
+
+from .. import _C
+from .hip_ops import DropoutFn, EmbeddingFn, LinearFn, LstmLayerFn
+
+
+class _LayerWorkspace:
+    def __init__(self, T: int, B: int, H: int, device):
+        bf, f32 = torch.bfloat16, torch.float32
+        self.T, self.B, self.H = T, B, H
+        self.gx = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
+        self.h_all = torch.zeros(T + 1, B, H, dtype=bf, device=device)
+        self.c_all = torch.zeros(T + 1, B, H, dtype=f32, device=device)
+        self.gates = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
+        self.dY = torch.zeros(T, B, H, dtype=bf, device=device)
+        self.dG = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
+        self.dh_rec = torch.zeros(B, H, dtype=f32, device=device)
+        self.dc = torch.zeros(B, H, dtype=f32, device=device)
+
+
+class _LayerRuntime:
+    """Shadows + workspace for one LSTM layer."""
+
+    def __init__(self, layer, device):
+        self.H = layer.hidden_size
+        self.Hin = layer.input_size
+        bf = torch.bfloat16
+        self.Wx = torch.empty_like(layer.W_x, dtype=bf, device=device)
+        self.Wh = torch.empty_like(layer.W_h, dtype=bf, device=device)
+        self.WxT = torch.empty(self.Hin, 4 * self.H, dtype=bf, device=device)
+        self.WhT = torch.empty(self.H, 4 * self.H, dtype=bf, device=device)
+        self.bias_sum = torch.empty(4 * self.H, dtype=torch.float32,
+                                    device=device)
+        self.ws: Optional[_LayerWorkspace] = None
+
+    def refresh(self, layer, e):
+        self.Wx.copy_(layer.W_x)
+        self.Wh.copy_(layer.W_h)
+        e.transpose_bf16(self.Wx, self.WxT)
+        e.transpose_bf16(self.Wh, self.WhT)
+        torch.add(layer.b_x, layer.b_h, out=self.bias_sum)
+
+    def ensure_ws(self, T, B, device):
+        if self.ws is None or self.ws.T != T or self.ws.B != B:
+            self.ws = _LayerWorkspace(T, B, self.H, device)
+        return self.ws
+
+
+class _FcRuntime:
+    def __init__(self, fc, device):
+        bf = torch.bfloat16
+        V, H = fc.W.shape
+        self.W = torch.empty(V, H, dtype=bf, device=device)
+        self.WT = torch.empty(H, V, dtype=bf, device=device)
+
+    def refresh(self, fc, e):
+        self.W.copy_(fc.W)
+        e.transpose_bf16(self.W, self.WT)
+
+
+class HipModel:
+    def __init__(self, model):
+        self.model = model
+        self.device = next(model.parameters()).device
+        if self.device.type != "cuda":
+            raise RuntimeError("HipModel requires a ROCm GPU device")
+        self.e = _C.ext()
+        self.compute_dtype = torch.bfloat16
+        dev = self.device
+        self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,
+                                      device=dev)
+        self.layers = [_LayerRuntime(l, dev) for l in model.rnns]
+        self.fc = _FcRuntime(model.fc, dev)
+        self.dropout_seed = secrets.randbits(63)
+        self.dropout_counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
+        self._shadows_fresh = False
+
+    def set_compute_dtype(self, dtype):
+        if dtype != torch.bfloat16:
+            raise ValueError("HIP engine currently runs bf16 compute "
+                             "(fp32 master weights); use --engine eager for "
+                             "full-fp32 debugging")
+        self.compute_dtype = dtype
+
+    # ------------------------------------------------------------------
+    def refresh_shadows(self):
+        m = self.model
+        self.emb_W.copy_(m.embed.W)
+        for rt, layer in zip(self.layers, m.rnns):
+            rt.refresh(layer, self.e)
+        self.fc.refresh(m.fc, self.e)
+        self._shadows_fresh = True
+
+    def invalidate_shadows(self):
+        self._shadows_fresh = False
+
+    # ------------------------------------------------------------------
+    def forward(self, x, states, training: bool):
+        if not self._shadows_fresh:
+            self.refresh_shadows()
+        m = self.model
+        T, B = x.shape
+        p = m.dropout_p
+        idx = x.reshape(-1)
+        emb = EmbeddingFn.apply(m.embed.W, idx, self.emb_W)
+        cur = emb.view(T, B, m.hidden_size)
+        if training and p > 0:
+            cur = DropoutFn.apply(cur, p, self.dropout_seed,
+                                  self.dropout_counter)
+        new_states = list(states)
+        for i, (rt, layer) in enumerate(zip(self.layers, m.rnns)):
+            rt.ensure_ws(T, B, self.device)
+            h0, c0 = states[i]
+            out, hT, cT = LstmLayerFn.apply(cur, h0, c0, layer.W_x,
+                                            layer.W_h, layer.b_x, layer.b_h,
+                                            rt)
+            new_states[i] = (hT, cT)
+            cur = out
+            if training and p > 0:
+                cur = DropoutFn.apply(cur, p, self.dropout_seed,
+                                      self.dropout_counter)
+        scores = LinearFn.apply(cur.reshape(T * B, m.hidden_size), m.fc.W,
+                                m.fc.b, self.fc)
+        for i in range(len(new_states)):
+            states[i] = new_states[i]
+        return scores, states
+
+    # ------------------------------------------------------------------
+    def clip_and_sgd(self, lr: float, max_norm: float,
+                     grad_scale: float = 1.0):
+        """Fused global grad-norm clip + SGD (reference main.py:115-117).
+
+        norm^2 accumulated across every param grad on device; each
+        parameter's update also rewrites its bf16 shadow; transposed
+        shadows + folded bias refreshed afterwards. Returns the pre-clip
+        grad norm as a 0-d device tensor (sync only when printed).
+        """
+        m = self.model
+        e = self.e
+        self.norm2.zero_()
+        params = [p for p in m.parameters() if p.grad is not None]
+        for p in params:
+            g = p.grad
+            e.norm2_accum(g.reshape(-1), self.norm2)
+        shadow_of = {id(m.embed.W): self.emb_W, id(m.fc.W): self.fc.W}
+        for rt, layer in zip(self.layers, m.rnns):
+            shadow_of[id(layer.W_x)] = rt.Wx
+            shadow_of[id(layer.W_h)] = rt.Wh
+        for p in params:
+            sh = shadow_of.get(id(p))
+            e.sgd_update(p.data.view(-1), p.grad.reshape(-1),
+                         sh.view(-1) if sh is not None else None,
+                         self.norm2, max_norm, lr, grad_scale)
+        # refresh the derived shadows (transposes + folded biases)
+        for rt, layer in zip(self.layers, m.rnns):
+            e.transpose_bf16(rt.Wx, rt.WxT)
+            e.transpose_bf16(rt.Wh, rt.WhT)
+            torch.add(layer.b_x, layer.b_h, out=rt.bias_sum)
+        e.transpose_bf16(self.fc.W, self.fc.WT)
+        return (self.norm2.sqrt() * grad_scale).reshape(())

@@ -1,0 +1,188 @@
+"""Autograd wiring for the HIP kernel path.
+
+Each Function routes forward AND backward through the hand-written gfx950
+kernels (zaremba_amd/csrc/). Parameters enter as fp32 masters (autograd
+edges / checkpoint format); compute uses the bf16 shadow copies held by
+the HipModel runtime, and backward returns fp32 grads that accumulate
+into the masters' .grad (which the fused clip+SGD kernels consume).
+
+Workspace-aliasing contract: the per-layer LSTM workspaces (h_all, c_all,
+gates, dG, ...) are persistent and re-used every step — valid because the
+training loop is strictly fwd -> bwd -> step (one outstanding autograd
+graph), which is the reference's own structure (main.py:108-117).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import _C
+
+
+def ext():
+    return _C.ext()
+
+
+class EmbeddingFn(torch.autograd.Function):
+    """K1: gather fwd / fp32 scatter-add bwd (reference model.py:6-17)."""
+
+    @staticmethod
+    def forward(ctx, W_master, idx, shadow_W):
+        N = idx.numel()
+        H = W_master.size(1)
+        out = torch.empty(N, H, dtype=torch.bfloat16, device=idx.device)
+        ext().embedding_fwd(shadow_W, idx, out)
+        ctx.save_for_backward(idx)
+        ctx.V = W_master.size(0)
+        return out
+
+    @staticmethod
+    def backward(ctx, dY):
+        (idx,) = ctx.saved_tensors
+        dY = dY.contiguous()
+        dW = torch.zeros(ctx.V, dY.size(1), dtype=torch.float32,
+                         device=dY.device)
+        ext().embedding_bwd(dY, idx, dW)
+        return dW, None, None
+
+
+class DropoutFn(torch.autograd.Function):
+    """K5: philox inverted dropout; the mask is regenerated in backward
+    from (seed, saved_offset) — no mask tensor stored."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed, counter):
+        x = x.contiguous()
+        y = torch.empty_like(x)
+        saved_offset = torch.zeros(1, dtype=torch.int64, device=x.device)
+        ext().dropout_fwd(x, y, p, seed, counter, saved_offset)
+        ctx.p = p
+        ctx.seed = seed
+        ctx.save_for_backward(saved_offset)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (saved_offset,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.empty_like(dy)
+        ext().dropout_bwd(dy, dx, ctx.p, ctx.seed, saved_offset)
+        return dx, None, None, None
+
+
+class LinearFn(torch.autograd.Function):
+    """K6: output projection via the NT MFMA GEMM; scores fp32
+    (reference model.py:57-71)."""
+
+    @staticmethod
+    def forward(ctx, x, W_master, b_master, fc_rt):
+        # x: [N, H] bf16; fc_rt holds shadows W (bf16 [V,H]) and WT ([H,V])
+        x = x.contiguous()
+        N, H = x.shape
+        V = W_master.size(0)
+        scores = torch.empty(N, V, dtype=torch.float32, device=x.device)
+        ext().gemm(x, fc_rt.W, scores, b_master, False, False)
+        ctx.save_for_backward(x)
+        ctx.fc_rt = fc_rt
+        return scores
+
+    @staticmethod
+    def backward(ctx, dscores):
+        (x,) = ctx.saved_tensors
+        fc_rt = ctx.fc_rt
+        dsc = dscores.to(torch.bfloat16).contiguous()
+        N, H = x.shape
+        V = dsc.size(1)
+        dx = torch.empty(N, H, dtype=torch.bfloat16, device=x.device)
+        ext().gemm(dsc, fc_rt.WT, dx, None, False, False)
+        dW = torch.empty(V, H, dtype=torch.float32, device=x.device)
+        ext().gemm(dsc, x, dW, None, True, True)
+        db = dscores.sum(0)
+        return dx, dW, db, None
+
+
+class NllLossFn(torch.autograd.Function):
+    """K7: fused stable log-softmax + batch_size-scaled NLL
+    (reference main.py:77-84 semantics)."""
+
+    @staticmethod
+    def forward(ctx, scores, y, batch_size):
+        scores = scores.contiguous()
+        N, V = scores.shape
+        yflat = y.reshape(-1).contiguous()
+        lse = torch.empty(N, dtype=torch.float32, device=scores.device)
+        accum = torch.zeros(1, dtype=torch.float32, device=scores.device)
+        ext().lsm_nll_fwd(scores, yflat, lse, accum)
+        ctx.save_for_backward(scores, lse, yflat)
+        ctx.scale = float(batch_size) / N
+        return (accum * ctx.scale).reshape(())
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        scores, lse, yflat = ctx.saved_tensors
+        N, V = scores.shape
+        up = grad_out.to(torch.float32).reshape(1).contiguous()
+        dscores = torch.empty_like(scores)
+        ext().lsm_nll_bwd(scores, lse, yflat, up, ctx.scale, dscores)
+        return dscores, None, None
+
+
+def nll_loss_hip(scores, y, batch_size):
+    return NllLossFn.apply(scores, y, batch_size)
+
+
+class LstmLayerFn(torch.autograd.Function):
+    """K2-K4 + K8: one full LSTM layer unroll on the fused cell kernels.
+
+    forward: hoisted input gate GEMM (x@W_x^T + b_x + b_h, MFMA NT), then
+    the hipGraph-captured T-step fused-cell sequence.
+    backward: graph-captured reverse unroll (dgate elementwise + skinny
+    recurrent GEMM per step), then batched dW/dx GEMMs over the stacked
+    [T*B, .] buffers.
+    """
+
+    @staticmethod
+    def forward(ctx, x, h0, c0, Wx_m, Wh_m, bx_m, bh_m, rt):
+        # x: [T, B, H_in] bf16; rt: _LayerRuntime with shadows + workspaces
+        T, B, Hin = x.shape
+        H = rt.H
+        x2 = x.contiguous()
+        e = ext()
+        # input gate GEMM for the whole unroll: [T*B, 4H]
+        e.gemm(x2.view(T * B, Hin), rt.Wx, rt.ws.gx.view(T * B, 4 * H),
+               rt.bias_sum, False, False)
+        rt.ws.h_all[0].copy_(h0.to(torch.bfloat16))
+        rt.ws.c_all[0].copy_(c0.to(torch.float32))
+        e.lstm_seq_fwd(rt.ws.gx, rt.Wh, rt.ws.h_all, rt.ws.c_all, rt.ws.gates)
+        ctx.save_for_backward(x2)
+        ctx.rt = rt
+        out = rt.ws.h_all[1:]          # [T, B, H] bf16 view (aliases ws)
+        hT = rt.ws.h_all[T]
+        cT = rt.ws.c_all[T]
+        ctx.mark_non_differentiable(hT, cT)
+        return out, hT, cT
+
+    @staticmethod
+    def backward(ctx, dY, dhT, dcT):
+        (x2,) = ctx.saved_tensors
+        rt = ctx.rt
+        e = ext()
+        ws = rt.ws
+        T, B, H = ws.dY.shape
+        Hin = x2.size(2)
+        ws.dY.copy_(dY.to(torch.bfloat16))
+        ws.dc.zero_()
+        e.lstm_seq_bwd(ws.dY, ws.gates, ws.c_all, rt.WhT, ws.dG, ws.dh_rec,
+                       ws.dc)
+        TB = T * B
+        dG2 = ws.dG.view(TB, 4 * H)
+        # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x
+        dWh = torch.empty(4 * H, H, dtype=torch.float32, device=x2.device)
+        e.gemm(dG2, ws.h_all[:T].reshape(TB, H), dWh, None, True, True)
+        dWx = torch.empty(4 * H, Hin, dtype=torch.float32, device=x2.device)
+        e.gemm(dG2, x2.view(TB, Hin), dWx, None, True, True)
+        dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
+        e.gemm(dG2, rt.WxT, dx, None, False, False)
+        db = dG2.float().sum(0)  # grads of b_x and b_h are identical
+        return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db.clone(),
+                None)

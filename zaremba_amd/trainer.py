@@ -29,7 +29,14 @@ from .ops import functional as F_ref
 
 
 def nll_loss(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
-    """Stable batch_size-scaled NLL (reference main.py:77-84 semantics)."""
+    """Stable batch_size-scaled NLL (reference main.py:77-84 semantics).
+
+    Dispatch: fused HIP log-softmax+NLL kernel on GPU scores, eager
+    torch math otherwise (CPU, or ZAREMBA_AMD_FORCE_EAGER=1)."""
+    from . import _C
+    if scores.is_cuda and not _C.force_eager_env():
+        from .ops.hip_ops import nll_loss_hip
+        return nll_loss_hip(scores, y, y.size(1))
     return F_ref.nll_loss(scores, y)
 
 
@@ -141,11 +148,13 @@ def train(
             loss.backward()
             if dp is not None:
                 dp.finalize_backward()
-            # Under DP each rank holds the SUM of per-rank grad sums; the
-            # global batch is world*B so the reference math (grads of a
-            # B-batch loss) is recovered by averaging across ranks.
-            norm = sgd_step(model, lr, max_norm,
-                            grad_scale=(1.0 / dp.world_size) if dp is not None else 1.0)
+            # Under DP each rank holds the SUM of per-rank grads after the
+            # all-reduce. The reference loss SUMS over the batch dimension
+            # (main.py:82-84), so summed grads are exactly the gradients
+            # the reference would compute at batch_size = world * B — the
+            # "global batch 160" semantics of BASELINE config 4. No
+            # averaging (grad_scale stays 1).
+            norm = sgd_step(model, lr, max_norm)
             if i % cadence == 0 and is_rank0:
                 toc = timeit.default_timer()
                 norm_v = float(norm)
