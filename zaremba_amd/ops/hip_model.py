@@ -53,6 +53,7 @@ class _LayerRuntime:
                                     device=device)
         self.ws: Optional[_LayerWorkspace] = None
 
+    @torch.no_grad()
     def refresh(self, layer, e):
         self.Wx.copy_(layer.W_x)
         self.Wh.copy_(layer.W_h)
@@ -73,6 +74,7 @@ class _FcRuntime:
         self.W = torch.empty(V, H, dtype=bf, device=device)
         self.WT = torch.empty(H, V, dtype=bf, device=device)
 
+    @torch.no_grad()
     def refresh(self, fc, e):
         self.W.copy_(fc.W)
         e.transpose_bf16(self.W, self.WT)
@@ -104,6 +106,7 @@ class HipModel:
         self.compute_dtype = dtype
 
     # ------------------------------------------------------------------
+    @torch.no_grad()
     def refresh_shadows(self):
         m = self.model
         self.emb_W.copy_(m.embed.W)
@@ -147,6 +150,7 @@ class HipModel:
         return scores, states
 
     # ------------------------------------------------------------------
+    @torch.no_grad()
     def clip_and_sgd(self, lr: float, max_norm: float,
                      grad_scale: float = 1.0):
         """Fused global grad-norm clip + SGD (reference main.py:115-117).
