@@ -87,7 +87,7 @@ def main():
 
     def one_step(x, y):
         nonlocal states
-        model.zero_grad(set_to_none=False)
+        model.zero_grad(set_to_none=(dp is None))
         if dp is not None:
             dp.zero_grad()
         states = model.detach(states)

@@ -222,10 +222,16 @@ __global__ void norm2_accum_kernel(const float* __restrict__ g, int64_t n,
                                    float* __restrict__ accum) {
   __shared__ float scratch[8];
   float acc = 0.f;
+  const int64_t n4 = n / 4;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    float v = g[i];
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  for (; i < n4; i += stride) {
+    float4 v = g4[i];
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (n & 3)) {
+    float v = g[n4 * 4 + threadIdx.x];
     acc += v * v;
   }
   acc = block_reduce(acc, scratch, 0.f,
@@ -252,12 +258,25 @@ __global__ void sgd_update_kernel(float* __restrict__ master,
   float norm = sqrtf(*norm2) * grad_scale;
   float coef = max_norm / (norm + 1e-6f);
   coef = fminf(coef, 1.f) * lr * grad_scale;
+  const int64_t n4 = n / 4;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    float v = master[i] - coef * grad[i];
-    master[i] = v;
-    if (shadow) shadow[i] = f2bf(v);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float4* m4 = reinterpret_cast<float4*>(master);
+  const float4* g4 = reinterpret_cast<const float4*>(grad);
+  bf16x4* s4 = reinterpret_cast<bf16x4*>(shadow);
+  for (; i < n4; i += stride) {
+    float4 m = m4[i];
+    float4 g = g4[i];
+    m.x -= coef * g.x; m.y -= coef * g.y;
+    m.z -= coef * g.z; m.w -= coef * g.w;
+    m4[i] = m;
+    if (shadow) s4[i] = bf16x4{f2bf(m.x), f2bf(m.y), f2bf(m.z), f2bf(m.w)};
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (n & 3)) {
+    int64_t j = n4 * 4 + threadIdx.x;
+    float v = master[j] - coef * grad[j];
+    master[j] = v;
+    if (shadow) shadow[j] = f2bf(v);
   }
 }
 
@@ -270,28 +289,76 @@ void launch_sgd_update(float* master, const float* grad, bf16* shadow,
                      master, grad, shadow, norm2, max_norm, lr, grad_scale, n);
 }
 
-// transposed bf16 shadow refresh: dst[c][r] = src[r][c] (LDS-tiled)
+// transposed bf16 shadow refresh: dst[c][r] = src[r][c].
+// 64x64 LDS tile, 16-B vector loads AND stores; interior blocks take a
+// guard-free path (per-element guards around global loads serialize).
 __global__ void transpose_bf16_kernel(const bf16* __restrict__ src,
                                       bf16* __restrict__ dst, int R, int C) {
-  __shared__ bf16 tile[32][33];
-  int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
-  int tc = threadIdx.x % 32, tr = threadIdx.x / 32;  // 32x8 threads
-  for (int i = 0; i < 32; i += 8) {
-    int r = r0 + tr + i, c = c0 + tc;
-    tile[tr + i][tc] = (r < R && c < C) ? src[(int64_t)r * C + c] : (bf16)0.f;
+  __shared__ bf16 tile[64][72];  // +8 bf16 row pad (16 B): conflict relief
+  const int c0 = blockIdx.x * 64, r0 = blockIdx.y * 64;
+  const int t = threadIdx.x;
+  const int lr = t / 8;           // 0..31 (row within pass)
+  const int lc8 = (t % 8) * 8;    // 0..56
+  const bool interior = (r0 + 64 <= R) && (c0 + 64 <= C);
+  if (interior) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int r = lr + p * 32;
+      *reinterpret_cast<bf16x8*>(&tile[r][lc8]) =
+          *reinterpret_cast<const bf16x8*>(src + (int64_t)(r0 + r) * C + c0 +
+                                           lc8);
+    }
+  } else {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int r = lr + p * 32;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int rr = r0 + r, cc = c0 + lc8 + e;
+        tile[r][lc8 + e] =
+            (rr < R && cc < C) ? src[(int64_t)rr * C + cc] : (bf16)0.f;
+      }
+    }
   }
   __syncthreads();
-  for (int i = 0; i < 32; i += 8) {
-    int r = r0 + tc, c = c0 + tr + i;  // write transposed, coalesced in r
-    if (r < R && c < C) dst[(int64_t)c * R + r] = tile[tc][tr + i];
+  // write: thread covers (col, 8 rows): dst row = c0+col, cols r0+r8..+8
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int c = lr + p * 32;     // source col == dst row offset
+    bf16x8 v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v[e] = tile[lc8 + e][c];
+    if (interior) {
+      *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + c) * R + r0 + lc8) = v;
+    } else if (c0 + c < C) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (r0 + lc8 + e < R) dst[(int64_t)(c0 + c) * R + r0 + lc8 + e] = v[e];
+    }
   }
 }
 
 void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
                            hipStream_t stream) {
-  dim3 grid(cdiv(C, 32), cdiv(R, 32));
+  dim3 grid(cdiv(C, 64), cdiv(R, 64));
   hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream, src,
                      dst, R, C);
+}
+
+// column sum of a bf16 matrix -> f32 (bias grads: db = colsum(dG))
+__global__ void colsum_bf16_kernel(const bf16* __restrict__ in,
+                                   float* __restrict__ out, int Rr, int Cc) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= Cc) return;
+  float acc = 0.f;
+  for (int r = 0; r < Rr; ++r) acc += bf2f(in[(int64_t)r * Cc + c]);
+  out[c] = acc;
+}
+
+void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(colsum_bf16_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
+                     stream, in, out, R, C);
 }
 
 }  // namespace zamd

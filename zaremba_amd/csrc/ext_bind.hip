@@ -294,6 +294,12 @@ static void sgd_update(torch::Tensor& master, const torch::Tensor& grad,
                     master.numel(), current_stream());
 }
 
+static void colsum_bf16(const torch::Tensor& in, torch::Tensor& out) {
+  int R = in.size(0), C = in.size(1);
+  TORCH_CHECK(out.numel() == C);
+  launch_colsum_bf16(bf_ptr(in), f_ptr_mut(out), R, C, current_stream());
+}
+
 static void transpose_bf16(const torch::Tensor& src, torch::Tensor& dst) {
   int R = src.size(0), C = src.size(1);
   TORCH_CHECK(dst.size(0) == C && dst.size(1) == R);
@@ -318,6 +324,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm2_accum", &zamd::norm2_accum);
   m.def("sgd_update", &zamd::sgd_update);
   m.def("transpose_bf16", &zamd::transpose_bf16);
+  m.def("colsum_bf16", &zamd::colsum_bf16);
   m.def("set_use_graphs", &zamd::set_use_graphs);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

@@ -3,40 +3,54 @@
 // (SURVEY.md §2.3 K3+K4, reference model.py:34-45 math; gate order i,f,o,n).
 //
 // Forward, one timestep (launched T times per layer by the C++ sequence
-// driver in ext.cpp, hipGraph-captured):
+// driver in ext_bind.hip, hipGraph-captured):
 //   gates = h_prev @ W_h^T + gx_t          (gx already holds x@W_x^T+b_x+b_h)
 //   i,f,o = sigmoid(g0,g1,g2); n = tanh(g3)
 //   c = f*c_prev + i*n ; h = o*tanh(c)
 //
-// Geometry (skinny-M design, B<=32): grid = ceil(H/16) workgroups, each
-// owning 16 hidden units; 4 waves per workgroup, wave g computes gate g's
-// [32(M) x 16] tile with v_mfma_f32_16x16x32_bf16 over the full K=H
-// reduction. h_prev is staged once into LDS ([32][Hpad] bf16); W_h B-
-// fragments are read straight from global (row-major [4H, H], K-contig —
-// W_h is L2/LLC-resident across the T-step unroll, which is the point of
-// the per-step relaunch design on 8 XCDs). The cell pointwise update +
-// state write happen in the same kernel via an LDS gate exchange.
+// Design (skinny-M, B<=32, measured on MI355X):
+//   * grid = ceil(H/16) workgroups x 4 waves; wave g owns gate g's
+//     [32 x 16] tile and reduces the full K=H with
+//     v_mfma_f32_16x16x32_bf16, fp32 accumulation.
+//   * No LDS staging: h_prev (60 KB) and the W_h slice are L2/LLC-
+//     resident across the T-step unroll (the point of the per-step
+//     relaunch design), so both MFMA operands are loaded straight from
+//     global with 16-B fragment reads. Out-of-range rows are CLAMPED to
+//     a valid row (finite garbage that only lands in discarded output
+//     rows/cols); only the K-tail step is element-guarded, because a
+//     zero A-fragment against uninitialized W bytes could make 0*NaN.
+//   * Depth-4 software prefetch on the fragment loads — per-element
+//     bounds branches or unprefetched chains leave the wave latency-
+//     bound at ~900 cycles/step (measured 26 us/step before; see
+//     profiles/).
+//   * The cell pointwise update + state write happen in-kernel through
+//     an 8 KB LDS gate exchange. c carries in fp32 across the epoch
+//     (truncated-BPTT state, reference main.py:110-111); h is bf16.
 //
-// c is carried in fp32 across the whole epoch (truncated-BPTT state,
-// reference main.py:110-111); h is bf16 (it feeds GEMMs).
-//
-// Backward, one timestep, two kernels (sequence driver loops t=T-1..0):
+// Backward, one timestep, two kernels (driver loops t=T-1..0):
 //   lstm_cell_bwd_elt: dgates_t from (dy_t + dh_rec, dc), updates dc.
-//   smallm_gemm_nt:    dh_rec = dgates_t @ W_h  (via the W_h^T shadow,
-//                      K-contiguous; 4 waves split K, LDS reduce).
+//   smallm_gemm_nt:    dh_rec = dgates_t @ W_h  (via the W_h^T shadow;
+//                      4 waves round-robin the K steps, LDS reduce).
 #include "common.h"
 
 namespace zamd {
 
 constexpr int CELL_THREADS = 256;
 
+// Guarded fragment load for the K-tail: elements past `limit` are zero.
+DEV_INLINE bf16x8 frag_tail(const bf16* p, int k, int limit) {
+  bf16x8 v = {};
+#pragma unroll
+  for (int e = 0; e < 8; ++e)
+    if (k + e < limit) v[e] = p[k + e];
+  return v;
+}
+
 // ---------------------------------------------------------------------------
 // Forward cell
 // ---------------------------------------------------------------------------
-// LDS budget (H=1500): h tile 32 * (ceil(H/32)*32 + 8) * 2B = 96.8 KB,
-// plus the 4x32x16 fp32 gate-exchange buffer (8 KB) -> one block per CU.
-template <int MAXB>  // padded batch rows (32)
-__global__ __launch_bounds__(CELL_THREADS, 1) void lstm_cell_fwd_kernel(
+template <int MAXB>
+__global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
     const bf16* __restrict__ h_prev,   // [B, H]
     const float* __restrict__ c_prev,  // [B, H]
     const bf16* __restrict__ gx,       // [B, 4H] this timestep's input gates
@@ -45,11 +59,7 @@ __global__ __launch_bounds__(CELL_THREADS, 1) void lstm_cell_fwd_kernel(
     float* __restrict__ c_out,         // [B, H]
     bf16* __restrict__ gates_out,      // [B, 4H] post-activation i,f,o,n
     int B, int H) {
-  const int KSTEPS = (H + 31) / 32;
-  const int HPAD = KSTEPS * 32 + 8;  // +8 bf16 row pad: conflict-free b128 reads
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* hs = reinterpret_cast<bf16*>(smem);                  // [MAXB][HPAD]
-  float* gbuf = reinterpret_cast<float*>(smem + MAXB * HPAD * 2);  // [4][MAXB][16]
+  __shared__ float gbuf[4 * MAXB * 16];
 
   const int j0 = blockIdx.x * 16;  // hidden-unit slice
   const int g = wave_id();         // gate index (i,f,o,n)
@@ -57,71 +67,70 @@ __global__ __launch_bounds__(CELL_THREADS, 1) void lstm_cell_fwd_kernel(
   const int lm = l & 15;
   const int lk = (l >> 4) * 8;
 
-  // ---- stage h_prev -> LDS (guarded vec8 loads, zero fill tails) ----------
-  {
-    const int vec_per_row = HPAD / 8;  // HPAD % 8 == 0
-    for (int idx = threadIdx.x; idx < MAXB * vec_per_row;
-         idx += CELL_THREADS) {
-      int b = idx / vec_per_row;
-      int k = (idx % vec_per_row) * 8;
-      bf16x8 v = {};
-      if (b < B && k < H) {
-        const bf16* p = h_prev + (int64_t)b * H + k;
-        if (k + 8 <= H) {
-          v = *reinterpret_cast<const bf16x8*>(p);
-        } else {
+  // Row-clamped operand base pointers (clamp => finite garbage that only
+  // reaches discarded outputs).
+  const int a0r = lm < B ? lm : B - 1;
+  const int a1r = (16 + lm) < B ? (16 + lm) : B - 1;
+  const int wr = g * H + (j0 + lm < H ? j0 + lm : H - 1);
+  const bf16* pa0 = h_prev + (int64_t)a0r * H;
+  const bf16* pa1 = h_prev + (int64_t)a1r * H;
+  const bf16* pw = W_h + (int64_t)wr * H;
+
+  const int full = H / 32;         // unguarded 32-k steps
+  const bool tail = (full * 32) < H;
+
+  f32x4 acc0 = {}, acc1 = {};
+
+  // 8-step chunks: 24 x 16-B loads issued per iteration before any wait,
+  // amortizing L2 latency over 16 MFMAs (hipcc groups loads at the top of
+  // the iteration; cross-iteration register rotation does not survive the
+  // scheduler, so work WITH that shape instead).
+  int ks = 0;
+  for (; ks + 8 <= full; ks += 8) {
+    bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) v[e] = (k + e < H) ? p[e] : (bf16)0.f;
-        }
-      }
-      *reinterpret_cast<bf16x8*>(hs + (int64_t)b * HPAD + k) = v;
+    for (int u = 0; u < 8; ++u) {
+      a0v[u] = *reinterpret_cast<const bf16x8*>(pa0 + (ks + u) * 32 + lk);
+      a1v[u] = *reinterpret_cast<const bf16x8*>(pa1 + (ks + u) * 32 + lk);
+      bwv[u] = *reinterpret_cast<const bf16x8*>(pw + (ks + u) * 32 + lk);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      acc0 = mfma_16x16x32_bf16(a0v[u], bwv[u], acc0);
+      acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
     }
   }
-  __syncthreads();
-
-  // ---- per-gate MFMA reduction over K = H --------------------------------
-  f32x4 acc[MAXB / 16] = {};
-  const int wrow = g * H + j0 + lm;        // W_h row for this lane's column
-  const bool col_ok = (j0 + lm) < H;
-  const bf16* wp = W_h + (int64_t)wrow * H;
-  for (int ks = 0; ks < KSTEPS; ++ks) {
-    int k = ks * 32 + lk;
-    bf16x8 bfrag = {};
-    if (col_ok) {
-      if (k + 8 <= H) {
-        bfrag = *reinterpret_cast<const bf16x8*>(wp + k);
-      } else if (k < H) {
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          bfrag[e] = (k + e < H) ? wp[k + e] : (bf16)0.f;
-      }
-    }
-#pragma unroll
-    for (int mf = 0; mf < MAXB / 16; ++mf) {
-      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-          hs + (int64_t)(mf * 16 + lm) * HPAD + ks * 32 + lk);
-      acc[mf] = mfma_16x16x32_bf16(afrag, bfrag, acc[mf]);
-    }
+  for (; ks < full; ++ks) {
+    bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + ks * 32 + lk);
+    bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + ks * 32 + lk);
+    bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pw + ks * 32 + lk);
+    acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
+    acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
+  }
+  if (tail) {
+    const int k = full * 32 + lk;
+    bf16x8 a0t = frag_tail(pa0, k, H);
+    bf16x8 a1t = frag_tail(pa1, k, H);
+    bf16x8 bwt = frag_tail(pw, k, H);
+    acc0 = mfma_16x16x32_bf16(a0t, bwt, acc0);
+    acc1 = mfma_16x16x32_bf16(a1t, bwt, acc1);
   }
 
   // ---- exchange gate tiles through LDS -----------------------------------
   // C/D map: col = l&15, row = (l>>4)*4 + r.
   const int fr0 = (l >> 4) * 4;
 #pragma unroll
-  for (int mf = 0; mf < MAXB / 16; ++mf) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int row = mf * 16 + fr0 + r;
-      gbuf[(g * MAXB + row) * 16 + lm] = acc[mf][r];
-    }
+  for (int r = 0; r < 4; ++r) {
+    gbuf[(g * MAXB + fr0 + r) * 16 + lm] = acc0[r];
+    gbuf[(g * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
   }
   __syncthreads();
 
   // ---- pointwise cell update ---------------------------------------------
   for (int idx = threadIdx.x; idx < B * 16; idx += CELL_THREADS) {
-    int b = idx / 16;
-    int jj = idx % 16;
-    int j = j0 + jj;
+    const int b = idx / 16;
+    const int jj = idx % 16;
+    const int j = j0 + jj;
     if (j >= H) continue;
     const int64_t gbase = (int64_t)b * 4 * H + j;
     float gi = gbuf[(0 * MAXB + b) * 16 + jj] + bf2f(gx[gbase + 0 * H]);
@@ -147,33 +156,23 @@ void launch_lstm_cell_fwd(const bf16* h_prev, const float* c_prev,
                           const bf16* gx, const bf16* W_h, bf16* h_out,
                           float* c_out, bf16* gates_out, int B, int H,
                           hipStream_t stream) {
-  const int KSTEPS = (H + 31) / 32;
-  const int HPAD = KSTEPS * 32 + 8;
-  size_t lds = (size_t)32 * HPAD * 2 + 4 * 32 * 16 * sizeof(float);
-  int grid = cdiv(H, 16);
-  hipLaunchKernelGGL((lstm_cell_fwd_kernel<32>), dim3(grid),
-                     dim3(CELL_THREADS), lds, stream, h_prev, c_prev, gx, W_h,
+  hipLaunchKernelGGL((lstm_cell_fwd_kernel<32>), dim3(cdiv(H, 16)),
+                     dim3(CELL_THREADS), 0, stream, h_prev, c_prev, gx, W_h,
                      h_out, c_out, gates_out, B, H);
 }
 
 // ---------------------------------------------------------------------------
 // Backward: per-timestep elementwise dgate kernel
 // ---------------------------------------------------------------------------
-// dh_t = dy_t + dh_rec (recurrent grad from step t+1; fp32 buffer)
-// tc = tanh(c_t);  do = dh*tc;  dct = dc + dh*o*(1-tc^2)
-// di = dct*n; df = dct*c_prev; dn = dct*i; do as above
-// pre-activation: dgi = di*i*(1-i); dgf = df*f*(1-f); dgo = do*o*(1-o);
-//                 dgn = dn*(1-n^2)
-// carried: dc <- dct * f
+// dh_t = dy_t + dh_rec; tc = tanh(c_t); do = dh*tc
+// dct = dc + dh*o*(1-tc^2); di = dct*n; df = dct*c_prev; dn = dct*i
+// pre-activation grads via the sigmoid/tanh local derivatives; carried
+// dc <- dct * f.
 __global__ void lstm_cell_bwd_elt_kernel(
-    const bf16* __restrict__ dy,       // [B,H] upstream at step t (bf16)
-    const float* __restrict__ dh_rec,  // [B,H] recurrent grad (nullptr at t=T-1)
-    float* __restrict__ dc,            // [B,H] carried, updated in place
-    const bf16* __restrict__ gates,    // [B,4H] saved i,f,o,n
-    const float* __restrict__ c_prev,  // [B,H] c_{t-1}
-    const float* __restrict__ c_new,   // [B,H] c_t
-    bf16* __restrict__ dG,             // [B,4H] out: pre-activation grads
-    int B, int H) {
+    const bf16* __restrict__ dy, const float* __restrict__ dh_rec,
+    float* __restrict__ dc, const bf16* __restrict__ gates,
+    const float* __restrict__ c_prev, const float* __restrict__ c_new,
+    bf16* __restrict__ dG, int B, int H) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= B * H) return;
   int b = idx / H, j = idx % H;
@@ -209,104 +208,96 @@ void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
 // ---------------------------------------------------------------------------
 // Skinny-M NT GEMM: C[M<=32, N] (fp32) = A[M,K] bf16 @ B[N,K]^T bf16
 // ---------------------------------------------------------------------------
-// Used for the recurrent backward hop dh_rec = dG_t @ W_h (B = the W_h^T
-// shadow, [H, 4H] row-major = N x K with K contiguous). Grid = ceil(N/16);
-// 4 waves split K and reduce through LDS. A is staged in K-chunks.
-constexpr int SMK_CHUNK = 1024;  // K elements per staged A chunk
-
+// The recurrent backward hop dh_rec = dG_t @ W_h (B = the W_h^T shadow,
+// [H, 4H] row-major, K contiguous). Grid = ceil(N/16); the 4 waves
+// round-robin the 32-wide K steps (wave w takes steps w, w+4, ...) so
+// every wave sees uniform full steps; one LDS reduction at the end.
+// Same streaming design as the forward cell: row-clamped direct global
+// fragment loads, depth-4 prefetch, element guards only on the K tail.
+// Requires K % 8 == 0 (K = 4H here).
 template <int MAXB>
-__global__ __launch_bounds__(CELL_THREADS, 1) void smallm_gemm_nt_kernel(
+__global__ __launch_bounds__(CELL_THREADS) void smallm_gemm_nt_kernel(
     const bf16* __restrict__ A,  // [M, K]
     const bf16* __restrict__ B_, // [N, K]
     float* __restrict__ C,       // [M, N]
     int M, int N, int K) {
-  constexpr int CP = SMK_CHUNK + 8;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* as = reinterpret_cast<bf16*>(smem);                    // [MAXB][CP]
-  float* red = reinterpret_cast<float*>(smem + MAXB * CP * 2); // [4][MAXB][16]
+  __shared__ float red[4 * MAXB * 16];
 
   const int n0 = blockIdx.x * 16;
   const int w = wave_id();
   const int l = lane_id();
   const int lm = l & 15;
   const int lk = (l >> 4) * 8;
-  const bool col_ok = (n0 + lm) < N;
-  const bf16* bp = B_ + (int64_t)(n0 + lm) * K;
 
-  f32x4 acc[MAXB / 16] = {};
-  for (int k0 = 0; k0 < K; k0 += SMK_CHUNK) {
-    const int klen = min(SMK_CHUNK, K - k0);
-    // stage A chunk
-    {
-      const int vec = CP / 8;
-      for (int idx = threadIdx.x; idx < MAXB * vec; idx += CELL_THREADS) {
-        int b = idx / vec;
-        int k = (idx % vec) * 8;
-        bf16x8 v = {};
-        if (b < M && k < klen) {
-          const bf16* p = A + (int64_t)b * K + k0 + k;
-          if (k + 8 <= klen) {
-            v = *reinterpret_cast<const bf16x8*>(p);
-          } else {
+  const int a0r = lm < M ? lm : M - 1;
+  const int a1r = (16 + lm) < M ? (16 + lm) : M - 1;
+  const int br = n0 + lm < N ? n0 + lm : N - 1;
+  const bf16* pa0 = A + (int64_t)a0r * K;
+  const bf16* pa1 = A + (int64_t)a1r * K;
+  const bf16* pb = B_ + (int64_t)br * K;
+
+  const int nsteps = (K + 31) / 32;
+  const int full = K / 32;
+
+  f32x4 acc0 = {}, acc1 = {};
+
+  // wave w owns steps w, w+4, w+8, ... ; i-th owned step = w + 4i.
+  // 8-owned-step chunks (24 loads in flight per iteration).
+  const int nown = (full - w + 3) / 4;  // owned FULL steps
+  int i = 0;
+  for (; i + 8 <= nown; i += 8) {
+    bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
-            for (int e = 0; e < 8; ++e)
-              v[e] = (k + e < klen) ? p[e] : (bf16)0.f;
-          }
-        }
-        *reinterpret_cast<bf16x8*>(as + (int64_t)b * CP + k) = v;
-      }
+    for (int u = 0; u < 8; ++u) {
+      const int kk = (w + 4 * (i + u)) * 32 + lk;
+      a0v[u] = *reinterpret_cast<const bf16x8*>(pa0 + kk);
+      a1v[u] = *reinterpret_cast<const bf16x8*>(pa1 + kk);
+      bwv[u] = *reinterpret_cast<const bf16x8*>(pb + kk);
     }
-    __syncthreads();
-    // each wave reduces its quarter of the chunk
-    const int kq = SMK_CHUNK / 4;  // 256
-    const int kw0 = w * kq;
-    for (int ks = 0; ks < kq; ks += 32) {
-      int kc = kw0 + ks;          // within chunk
-      if (kc >= klen) break;
-      int kg = k0 + kc + lk;      // global k for the B fragment
-      bf16x8 bfrag = {};
-      if (col_ok && kg < K) {
-        if (kg + 8 <= K) {
-          bfrag = *reinterpret_cast<const bf16x8*>(bp + kg);
-        } else {
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            bfrag[e] = (kg + e < K) ? bp[kg + e] : (bf16)0.f;
-        }
-      }
-#pragma unroll
-      for (int mf = 0; mf < MAXB / 16; ++mf) {
-        bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            as + (int64_t)(mf * 16 + lm) * CP + kc + lk);
-        acc[mf] = mfma_16x16x32_bf16(afrag, bfrag, acc[mf]);
-      }
+    for (int u = 0; u < 8; ++u) {
+      acc0 = mfma_16x16x32_bf16(a0v[u], bwv[u], acc0);
+      acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
     }
-    __syncthreads();
+  }
+  for (; i < nown; ++i) {
+    const int kk = (w + 4 * i) * 32 + lk;
+    bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + kk);
+    bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + kk);
+    bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pb + kk);
+    acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
+    acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
+  }
+  // K tail step (K % 32 != 0), owned by wave (full % 4)
+  if (full < nsteps && w == (full % 4)) {
+    const int k = full * 32 + lk;
+    bf16x8 a0t = frag_tail(pa0, k, K);
+    bf16x8 a1t = frag_tail(pa1, k, K);
+    bf16x8 bt = frag_tail(pb, k, K);
+    acc0 = mfma_16x16x32_bf16(a0t, bt, acc0);
+    acc1 = mfma_16x16x32_bf16(a1t, bt, acc1);
   }
 
-  // reduce the 4 wave partials via LDS
   const int fr0 = (l >> 4) * 4;
 #pragma unroll
-  for (int mf = 0; mf < MAXB / 16; ++mf)
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      red[(w * MAXB + mf * 16 + fr0 + r) * 16 + lm] = acc[mf][r];
+  for (int r = 0; r < 4; ++r) {
+    red[(w * MAXB + fr0 + r) * 16 + lm] = acc0[r];
+    red[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+  }
   __syncthreads();
   for (int idx = threadIdx.x; idx < M * 16; idx += CELL_THREADS) {
-    int b = idx / 16, jj = idx % 16;
+    const int b = idx / 16, jj = idx % 16;
     if (n0 + jj >= N) continue;
-    float v = 0.f;
-#pragma unroll
-    for (int ww = 0; ww < 4; ++ww) v += red[(ww * MAXB + b) * 16 + jj];
+    float v = red[(0 * MAXB + b) * 16 + jj] + red[(1 * MAXB + b) * 16 + jj] +
+              red[(2 * MAXB + b) * 16 + jj] + red[(3 * MAXB + b) * 16 + jj];
     C[(int64_t)b * N + n0 + jj] = v;
   }
 }
 
 void launch_smallm_gemm_nt(const bf16* A, const bf16* B, float* C, int M,
                            int N, int K, hipStream_t stream) {
-  size_t lds = (size_t)32 * (SMK_CHUNK + 8) * 2 + 4 * 32 * 16 * sizeof(float);
   hipLaunchKernelGGL((smallm_gemm_nt_kernel<32>), dim3(cdiv(N, 16)),
-                     dim3(CELL_THREADS), lds, stream, A, B, C, M, N, K);
+                     dim3(CELL_THREADS), 0, stream, A, B, C, M, N, K);
 }
 
 }  // namespace zamd

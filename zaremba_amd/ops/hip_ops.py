@@ -90,13 +90,20 @@ class LinearFn(torch.autograd.Function):
     def backward(ctx, dscores):
         (x,) = ctx.saved_tensors
         fc_rt = ctx.fc_rt
+        e = ext()
         dsc = dscores.to(torch.bfloat16).contiguous()
         N, H = x.shape
         V = dsc.size(1)
         dx = torch.empty(N, H, dtype=torch.bfloat16, device=x.device)
-        ext().gemm(dsc, fc_rt.WT, dx, None, False, False)
+        e.gemm(dsc, fc_rt.WT, dx, None, False, False)
+        # dW = dsc^T @ x via explicit transposes + the fast NT kernel (the
+        # TN staging path is register-starved; measured 2-3x slower)
+        dscT = torch.empty(V, N, dtype=torch.bfloat16, device=x.device)
+        e.transpose_bf16(dsc, dscT)
+        xT = torch.empty(H, N, dtype=torch.bfloat16, device=x.device)
+        e.transpose_bf16(x, xT)
         dW = torch.empty(V, H, dtype=torch.float32, device=x.device)
-        ext().gemm(dsc, x, dW, None, True, True)
+        e.gemm(dscT, xT, dW, None, False, False)
         db = dscores.sum(0)
         return dx, dW, db, None
 
@@ -176,13 +183,21 @@ class LstmLayerFn(torch.autograd.Function):
                        ws.dc)
         TB = T * B
         dG2 = ws.dG.view(TB, 4 * H)
-        # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x
+        # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x —
+        # weight grads via explicit transposes + the fast NT kernel.
+        dGT = torch.empty(4 * H, TB, dtype=torch.bfloat16, device=x2.device)
+        e.transpose_bf16(dG2, dGT)
+        hpT = torch.empty(H, TB, dtype=torch.bfloat16, device=x2.device)
+        e.transpose_bf16(ws.h_all[:T].reshape(TB, H), hpT)
         dWh = torch.empty(4 * H, H, dtype=torch.float32, device=x2.device)
-        e.gemm(dG2, ws.h_all[:T].reshape(TB, H), dWh, None, True, True)
+        e.gemm(dGT, hpT, dWh, None, False, False)
+        xT = torch.empty(Hin, TB, dtype=torch.bfloat16, device=x2.device)
+        e.transpose_bf16(x2.view(TB, Hin), xT)
         dWx = torch.empty(4 * H, Hin, dtype=torch.float32, device=x2.device)
-        e.gemm(dG2, x2.view(TB, Hin), dWx, None, True, True)
+        e.gemm(dGT, xT, dWx, None, False, False)
         dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
         e.gemm(dG2, rt.WxT, dx, None, False, False)
-        db = dG2.float().sum(0)  # grads of b_x and b_h are identical
+        db = torch.empty(4 * H, dtype=torch.float32, device=x2.device)
+        e.colsum_bf16(dG2, db)  # grads of b_x and b_h are identical
         return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db.clone(),
                 None)

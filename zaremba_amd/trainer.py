@@ -139,7 +139,9 @@ def train(
             lr = lr / factor
         for i, (x, y) in enumerate(trn):
             total_words += x.numel()
-            model.zero_grad(set_to_none=False)
+            # set_to_none saves a fill + an accumulate add per param; the DP
+            # bucketer needs stable grad views, so keep zeroing there.
+            model.zero_grad(set_to_none=(dp is None))
             if dp is not None:
                 dp.zero_grad()
             states = model.detach(states)
