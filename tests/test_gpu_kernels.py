@@ -134,14 +134,20 @@ def test_lstm_cell_fwd_step(ext, B, H):
 def test_lstm_seq_fwd_matches_step_loop(ext):
     torch.manual_seed(5)
     T, B, H = 6, 20, 650
+    KS = (H + 31) // 32
     gx = (torch.randn(T, B, 4 * H, device=dev()) * 0.5).to(torch.bfloat16)
     W_h = (torch.randn(4 * H, H, device=dev()) * 0.02).to(torch.bfloat16)
+    WhP = torch.empty(((H + 15) // 16) * 4 * KS * 64 * 8, device=dev(),
+                      dtype=torch.bfloat16)
+    ext.pack_gated_w(W_h, WhP, H, 4, H)
     h_all = torch.zeros(T + 1, B, H, device=dev(), dtype=torch.bfloat16)
+    h_pack = torch.zeros(T + 1, KS * 2 * 64 * 8, device=dev(),
+                         dtype=torch.bfloat16)
     c_all = torch.zeros(T + 1, B, H, device=dev(), dtype=torch.float32)
     gates = torch.empty(T, B, 4 * H, device=dev(), dtype=torch.bfloat16)
     h_all[0] = (torch.randn(B, H, device=dev()) * 0.3).to(torch.bfloat16)
     c_all[0] = torch.randn(B, H, device=dev()) * 0.3
-    ext.lstm_seq_fwd(gx, W_h, h_all, c_all, gates)
+    ext.lstm_seq_fwd(gx, WhP, h_all, h_pack, c_all, gates)
     h = h_all[0]
     c = c_all[0].clone()
     for t in range(T):

@@ -100,44 +100,49 @@ struct GraphCache {
 static GraphCache g_fwd_graphs, g_bwd_graphs;
 static bool g_use_graphs = true;
 
-static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h, bf16* h_all,
-                              float* c_all, bf16* gates, int T, int B, int H,
+static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_pack,
+                              bf16* h_all, bf16* h_pack, float* c_all,
+                              bf16* gates, int T, int B, int H,
                               hipStream_t stream) {
   const int64_t hstep = (int64_t)B * H;
   const int64_t gstep = (int64_t)B * 4 * H;
+  const int64_t pstep = (int64_t)((H + 31) / 32) * 2 * 64 * 8;
+  launch_pack_a(h_all, h_pack, B, H, stream);  // slot 0 = h0
   for (int t = 0; t < T; ++t) {
-    launch_lstm_cell_fwd(h_all + t * hstep, c_all + t * hstep, gx + t * gstep,
-                         W_h, h_all + (t + 1) * hstep,
-                         c_all + (t + 1) * hstep, gates + t * gstep, B, H,
-                         stream);
+    launch_lstm_cell_fwd(h_pack + t * pstep, c_all + t * hstep,
+                         gx + t * gstep, W_pack, h_all + (t + 1) * hstep,
+                         h_pack + (t + 1) * pstep, c_all + (t + 1) * hstep,
+                         gates + t * gstep, B, H, stream);
   }
 }
 
-// h_all/c_all are [T+1, B, H] with slot 0 pre-filled with (h0, c0).
-static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
-                         torch::Tensor& h_all, torch::Tensor& c_all,
-                         torch::Tensor& gates) {
+// h_all/c_all are [T+1, B, H] with slot 0 pre-filled with (h0, c0);
+// h_pack is the zero-prefilled [T+1, KS*2*64*8] packed-h workspace.
+static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_pack,
+                         torch::Tensor& h_all, torch::Tensor& h_pack,
+                         torch::Tensor& c_all, torch::Tensor& gates) {
   int T = gx.size(0), B = gx.size(1);
   int H = h_all.size(2);
   TORCH_CHECK(gx.size(2) == 4 * H, "gx must be [T,B,4H]");
   const bf16* gxp = bf_ptr(gx);
-  const bf16* whp = bf_ptr(W_h);
+  const bf16* whp = bf_ptr(W_pack);
   bf16* hp = bf_ptr_mut(h_all);
+  bf16* hpk = bf_ptr_mut(h_pack);
   float* cp = f_ptr_mut(c_all);
   bf16* gp = bf_ptr_mut(gates);
   auto stream = current_stream();
   if (!g_use_graphs) {
-    lstm_seq_fwd_body(gxp, whp, hp, cp, gp, T, B, H, stream);
+    lstm_seq_fwd_body(gxp, whp, hp, hpk, cp, gp, T, B, H, stream);
     return;
   }
   std::vector<uintptr_t> key{(uintptr_t)gxp, (uintptr_t)whp, (uintptr_t)hp,
-                             (uintptr_t)cp, (uintptr_t)gp, (uintptr_t)T,
-                             (uintptr_t)B, (uintptr_t)H};
+                             (uintptr_t)hpk, (uintptr_t)cp, (uintptr_t)gp,
+                             (uintptr_t)T, (uintptr_t)B, (uintptr_t)H};
   auto it = g_fwd_graphs.cache.find(key);
   if (it == g_fwd_graphs.cache.end()) {
     hipStream_t cs = g_fwd_graphs.capture_stream();
     HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
-    lstm_seq_fwd_body(gxp, whp, hp, cp, gp, T, B, H, cs);
+    lstm_seq_fwd_body(gxp, whp, hp, hpk, cp, gp, T, B, H, cs);
     hipGraph_t graph;
     HIP_CHECK(hipStreamEndCapture(cs, &graph));
     hipGraphExec_t exec;
@@ -149,8 +154,9 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
 }
 
 static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
-                              const float* c_all, const bf16* W_h_T, bf16* dG,
-                              float* dh_rec, float* dc, int T, int B, int H,
+                              const float* c_all, const bf16* WT_pack,
+                              bf16* dG, bf16* dG_pack, float* dh_rec,
+                              float* dc, int T, int B, int H,
                               hipStream_t stream) {
   const int64_t hstep = (int64_t)B * H;
   const int64_t gstep = (int64_t)B * 4 * H;
@@ -158,40 +164,43 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
     launch_lstm_cell_bwd_elt(dY + t * hstep,
                              (t == T - 1) ? nullptr : dh_rec, dc,
                              gates + t * gstep, c_all + t * hstep,
-                             c_all + (t + 1) * hstep, dG + t * gstep, B, H,
-                             stream);
-    launch_smallm_gemm_nt(dG + t * gstep, W_h_T, dh_rec, B, H, 4 * H, stream);
+                             c_all + (t + 1) * hstep, dG + t * gstep, dG_pack,
+                             B, H, stream);
+    launch_smallm_packed_nt(dG_pack, WT_pack, dh_rec, B, H, 4 * H, stream);
   }
 }
 
 // dc must be zero-filled by the caller before each call; dG is [T,B,4H];
-// dh_rec is a [B,H] f32 workspace. W_h_T is the transposed shadow [H,4H].
+// dG_pack the zero-prefilled packed-dG workspace; dh_rec a [B,H] f32
+// workspace. WT_pack is the packed W_h^T shadow.
 static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
                          const torch::Tensor& c_all,
-                         const torch::Tensor& W_h_T, torch::Tensor& dG,
-                         torch::Tensor& dh_rec, torch::Tensor& dc) {
+                         const torch::Tensor& WT_pack, torch::Tensor& dG,
+                         torch::Tensor& dG_pack, torch::Tensor& dh_rec,
+                         torch::Tensor& dc) {
   int T = dY.size(0), B = dY.size(1), H = dY.size(2);
   const bf16* dyp = bf_ptr(dY);
   const bf16* gp = bf_ptr(gates);
   const float* cp = f_ptr(c_all);
-  const bf16* wtp = bf_ptr(W_h_T);
+  const bf16* wtp = bf_ptr(WT_pack);
   bf16* dgp = bf_ptr_mut(dG);
+  bf16* dgpk = bf_ptr_mut(dG_pack);
   float* dhp = f_ptr_mut(dh_rec);
   float* dcp = f_ptr_mut(dc);
   auto stream = current_stream();
   if (!g_use_graphs) {
-    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dhp, dcp, T, B, H, stream);
+    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dgpk, dhp, dcp, T, B, H, stream);
     return;
   }
   std::vector<uintptr_t> key{(uintptr_t)dyp, (uintptr_t)gp, (uintptr_t)cp,
-                             (uintptr_t)wtp, (uintptr_t)dgp, (uintptr_t)dhp,
-                             (uintptr_t)dcp, (uintptr_t)T, (uintptr_t)B,
-                             (uintptr_t)H};
+                             (uintptr_t)wtp, (uintptr_t)dgp, (uintptr_t)dgpk,
+                             (uintptr_t)dhp, (uintptr_t)dcp, (uintptr_t)T,
+                             (uintptr_t)B, (uintptr_t)H};
   auto it = g_bwd_graphs.cache.find(key);
   if (it == g_bwd_graphs.cache.end()) {
     hipStream_t cs = g_bwd_graphs.capture_stream();
     HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
-    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dhp, dcp, T, B, H, cs);
+    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dgpk, dhp, dcp, T, B, H, cs);
     hipGraph_t graph;
     HIP_CHECK(hipStreamEndCapture(cs, &graph));
     hipGraphExec_t exec;
@@ -210,16 +219,43 @@ static void clear_graphs() {
   g_bwd_graphs.cache.clear();
 }
 
-// single-step cell entry points (used by unit tests)
+// single-step cell entry points (used by unit tests; pack on the fly)
+static torch::Tensor pack_a_tmp(const torch::Tensor& A) {
+  int B = A.size(0), K = A.size(1);
+  int KS = (K + 31) / 32;
+  auto out = torch::zeros({(int64_t)KS * 2 * 64 * 8}, A.options());
+  launch_pack_a(bf_ptr(A), bf_ptr_mut(out), B, K, current_stream());
+  return out;
+}
+
+static torch::Tensor pack_w_tmp(const torch::Tensor& W, int rows, int ngates,
+                                int K) {
+  int KS = (K + 31) / 32;
+  int nb = (rows + 15) / 16;
+  auto out = torch::empty({(int64_t)nb * ngates * KS * 64 * 8, }, W.options());
+  launch_pack_gated_w(bf_ptr(W), bf_ptr_mut(out), rows, ngates, K,
+                      current_stream());
+  return out;
+}
+
+static void pack_gated_w(const torch::Tensor& W, torch::Tensor& out,
+                         int64_t rows, int64_t ngates, int64_t K) {
+  launch_pack_gated_w(bf_ptr(W), bf_ptr_mut(out), (int)rows, (int)ngates,
+                      (int)K, current_stream());
+}
+
 static void lstm_cell_fwd_step(const torch::Tensor& h_prev,
                                const torch::Tensor& c_prev,
                                const torch::Tensor& gx,
                                const torch::Tensor& W_h, torch::Tensor& h_out,
                                torch::Tensor& c_out, torch::Tensor& gates) {
   int B = h_prev.size(0), H = h_prev.size(1);
-  launch_lstm_cell_fwd(bf_ptr(h_prev), f_ptr(c_prev), bf_ptr(gx), bf_ptr(W_h),
-                       bf_ptr_mut(h_out), f_ptr_mut(c_out), bf_ptr_mut(gates),
-                       B, H, current_stream());
+  auto hp = pack_a_tmp(h_prev);
+  auto wp = pack_w_tmp(W_h, H, 4, H);
+  auto hpo = torch::zeros_like(hp);
+  launch_lstm_cell_fwd(bf_ptr(hp), f_ptr(c_prev), bf_ptr(gx), bf_ptr(wp),
+                       bf_ptr_mut(h_out), bf_ptr_mut(hpo), f_ptr_mut(c_out),
+                       bf_ptr_mut(gates), B, H, current_stream());
 }
 
 static void smallm_gemm_nt(const torch::Tensor& A, const torch::Tensor& B,
@@ -227,8 +263,10 @@ static void smallm_gemm_nt(const torch::Tensor& A, const torch::Tensor& B,
   int M = A.size(0), K = A.size(1), N = B.size(0);
   TORCH_CHECK(B.size(1) == K && C.size(0) == M && C.size(1) == N);
   TORCH_CHECK(M <= 32, "smallm gemm requires M <= 32");
-  launch_smallm_gemm_nt(bf_ptr(A), bf_ptr(B), f_ptr_mut(C), M, N, K,
-                        current_stream());
+  auto ap = pack_a_tmp(A);
+  auto wp = pack_w_tmp(B, N, 1, K);
+  launch_smallm_packed_nt(bf_ptr(ap), bf_ptr(wp), f_ptr_mut(C), M, N, K,
+                          current_stream());
 }
 
 // ---------------------------------------------------------------------------
@@ -314,6 +352,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_fwd", &zamd::lstm_seq_fwd);
   m.def("lstm_seq_bwd", &zamd::lstm_seq_bwd);
   m.def("lstm_cell_fwd_step", &zamd::lstm_cell_fwd_step);
+  m.def("pack_gated_w", &zamd::pack_gated_w);
   m.def("smallm_gemm_nt", &zamd::smallm_gemm_nt);
   m.def("embedding_fwd", &zamd::embedding_fwd);
   m.def("embedding_bwd", &zamd::embedding_bwd);

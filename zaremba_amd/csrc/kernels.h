@@ -16,17 +16,22 @@ void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
                    int M, int N, int K, int lda, int ldb, int ldc,
                    hipStream_t stream);
 
-// lstm.hip
-void launch_lstm_cell_fwd(const bf16* h_prev, const float* c_prev,
-                          const bf16* gx, const bf16* W_h, bf16* h_out,
-                          float* c_out, bf16* gates_out, int B, int H,
-                          hipStream_t stream);
+// lstm.hip (fragment-packed operands; see lstm.hip header comment)
+void launch_pack_gated_w(const bf16* W, bf16* out, int rows, int ngates,
+                         int K, hipStream_t stream);
+void launch_pack_a(const bf16* A, bf16* out, int B, int K,
+                   hipStream_t stream);
+void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
+                          const bf16* gx, const bf16* W_pack, bf16* h_out,
+                          bf16* h_pack_out, float* c_out, bf16* gates_out,
+                          int B, int H, hipStream_t stream);
 void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
                               const bf16* gates, const float* c_prev,
-                              const float* c_new, bf16* dG, int B, int H,
-                              hipStream_t stream);
-void launch_smallm_gemm_nt(const bf16* A, const bf16* B, float* C, int M,
-                           int N, int K, hipStream_t stream);
+                              const float* c_new, bf16* dG, bf16* dG_pack,
+                              int B, int H, hipStream_t stream);
+void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream);
 
 // elementwise.hip
 void launch_embedding_fwd(const bf16* W, const int64_t* idx, bf16* out,

@@ -8,54 +8,110 @@
 //   i,f,o = sigmoid(g0,g1,g2); n = tanh(g3)
 //   c = f*c_prev + i*n ; h = o*tanh(c)
 //
-// Design (skinny-M, B<=32, measured on MI355X):
-//   * grid = ceil(H/16) workgroups x 4 waves; wave g owns gate g's
-//     [32 x 16] tile and reduces the full K=H with
-//     v_mfma_f32_16x16x32_bf16, fp32 accumulation.
-//   * No LDS staging: h_prev (60 KB) and the W_h slice are L2/LLC-
-//     resident across the T-step unroll (the point of the per-step
-//     relaunch design), so both MFMA operands are loaded straight from
-//     global with 16-B fragment reads. Out-of-range rows are CLAMPED to
-//     a valid row (finite garbage that only lands in discarded output
-//     rows/cols); only the K-tail step is element-guarded, because a
-//     zero A-fragment against uninitialized W bytes could make 0*NaN.
-//   * Depth-4 software prefetch on the fragment loads — per-element
-//     bounds branches or unprefetched chains leave the wave latency-
-//     bound at ~900 cycles/step (measured 26 us/step before; see
-//     profiles/).
-//   * The cell pointwise update + state write happen in-kernel through
-//     an 8 KB LDS gate exchange. c carries in fp32 across the epoch
-//     (truncated-BPTT state, reference main.py:110-111); h is bf16.
+// Fragment-packed operand design (MI355X-specific, measured):
+//   The natural MFMA fragment access for a skinny [B<=32, H] x [H, 4H]
+//   GEMM scatters every 16-B lane load across 16 weight rows — each
+//   wave instruction becomes ~16 separate 64-B L2 transactions, and the
+//   kernel is transaction-rate-bound (~17 us/step). Instead, both
+//   operands are kept in FRAGMENT-PACKED layout ([kstep][frag][lane][8]
+//   bf16), so every wave load is one contiguous 1 KB burst:
+//     * W_h / W_h^T are packed once per SGD step (pack_gated_w kernel,
+//       reading the bf16 shadows) — amortized over the 70 cell launches
+//       of every training step,
+//     * h is written in packed layout by the PREVIOUS cell step's
+//       epilogue (dual store), and h0 by a tiny pack_a kernel at
+//       sequence start,
+//     * dgates are written packed by the backward elementwise kernel for
+//       the recurrent-hop GEMM that follows it.
+//   Pad rows (b >= B) and K-tail slots are zero in the packed buffers
+//   (zero-prefilled workspaces / zero-filled packing), so the inner
+//   loops are completely uniform — no bounds branches, no frag guards.
 //
-// Backward, one timestep, two kernels (driver loops t=T-1..0):
-//   lstm_cell_bwd_elt: dgates_t from (dy_t + dh_rec, dc), updates dc.
-//   smallm_gemm_nt:    dh_rec = dgates_t @ W_h  (via the W_h^T shadow;
-//                      4 waves round-robin the K steps, LDS reduce).
+// Packed layout: frag f of kstep ks, lane l, elem e lives at
+//   ((ks*2 + f)*64 + l)*8 + e          (A operands: 2 M-fragments)
+//   (((blk*NG + g)*KS + ks)*64 + l)*8 + e   (W operands, NG gates)
+// with the standard v_mfma_f32_16x16x32_bf16 maps
+//   A: row = (l&15) + 16*f, k = ks*32 + (l>>4)*8 + e
+//   B: col = l&15,          k = ks*32 + (l>>4)*8 + e.
 #include "common.h"
 
 namespace zamd {
 
 constexpr int CELL_THREADS = 256;
 
-// Guarded fragment load for the K-tail: elements past `limit` are zero.
-DEV_INLINE bf16x8 frag_tail(const bf16* p, int k, int limit) {
+// ---------------------------------------------------------------------------
+// Packing kernels
+// ---------------------------------------------------------------------------
+// W[N-space, K] (row-major, k-contiguous; N-space = ngates * rows) ->
+// packed [ceil(rows/16)][ngates][ceil(K/32)][64][8], zero-filled pads.
+__global__ void pack_gated_w_kernel(const bf16* __restrict__ W,
+                                    bf16* __restrict__ out, int rows,
+                                    int ngates, int K) {
+  const int KS = (K + 31) / 32;
+  const int nb = (rows + 15) / 16;
+  const int64_t total = (int64_t)nb * ngates * KS * 64;
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= total) return;
+  const int l = idx & 63;
+  const int ks = (idx >> 6) % KS;
+  const int g = (int)((idx >> 6) / KS) % ngates;
+  const int bj = (int)((idx >> 6) / KS / ngates);
+  const int row = bj * 16 + (l & 15);
+  const int k = ks * 32 + (l >> 4) * 8;
   bf16x8 v = {};
+  if (row < rows) {
+    const bf16* p = W + ((int64_t)g * rows + row) * K + k;
+    if (k + 8 <= K) {
+      v = *reinterpret_cast<const bf16x8*>(p);
+    } else {
 #pragma unroll
-  for (int e = 0; e < 8; ++e)
-    if (k + e < limit) v[e] = p[k + e];
-  return v;
+      for (int e = 0; e < 8; ++e) v[e] = (k + e < K) ? p[e] : (bf16)0.f;
+    }
+  }
+  reinterpret_cast<bf16x8*>(out)[idx] = v;
+}
+
+void launch_pack_gated_w(const bf16* W, bf16* out, int rows, int ngates,
+                         int K, hipStream_t stream) {
+  const int KS = (K + 31) / 32;
+  const int nb = (rows + 15) / 16;
+  int64_t total = (int64_t)nb * ngates * KS * 64;
+  hipLaunchKernelGGL(pack_gated_w_kernel, dim3(cdiv(total, 256)), dim3(256),
+                     0, stream, W, out, rows, ngates, K);
+}
+
+// A[B, K] -> packed [ceil(K/32)][2][64][8]; pad rows/K-tail slots must
+// already be zero in `out` (zero-prefilled persistent workspace).
+__global__ void pack_a_kernel(const bf16* __restrict__ A,
+                              bf16* __restrict__ out, int B, int K) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (int64_t)B * K) return;
+  const int b = (int)(idx / K);
+  const int k = (int)(idx % K);
+  const int ks = k / 32, sub = k % 32;
+  const int l = (b & 15) + 16 * (sub / 8);
+  const int e = sub % 8;
+  const int f = b / 16;
+  out[(((int64_t)ks * 2 + f) * 64 + l) * 8 + e] = A[idx];
+}
+
+void launch_pack_a(const bf16* A, bf16* out, int B, int K,
+                   hipStream_t stream) {
+  hipLaunchKernelGGL(pack_a_kernel, dim3(cdiv((int64_t)B * K, 256)),
+                     dim3(256), 0, stream, A, out, B, K);
 }
 
 // ---------------------------------------------------------------------------
-// Forward cell
+// Forward cell (packed operands)
 // ---------------------------------------------------------------------------
 template <int MAXB>
 __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
-    const bf16* __restrict__ h_prev,   // [B, H]
+    const bf16* __restrict__ h_pack,   // [KS][2][64][8] packed h_{t}
     const float* __restrict__ c_prev,  // [B, H]
     const bf16* __restrict__ gx,       // [B, 4H] this timestep's input gates
-    const bf16* __restrict__ W_h,      // [4H, H] row-major
+    const bf16* __restrict__ W_pack,   // [nb][4][KS][64][8]
     bf16* __restrict__ h_out,          // [B, H]
+    bf16* __restrict__ h_pack_out,     // packed h_{t+1}
     float* __restrict__ c_out,         // [B, H]
     bf16* __restrict__ gates_out,      // [B, 4H] post-activation i,f,o,n
     int B, int H) {
@@ -65,34 +121,21 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
   const int g = wave_id();         // gate index (i,f,o,n)
   const int l = lane_id();
   const int lm = l & 15;
-  const int lk = (l >> 4) * 8;
+  const int KS = (H + 31) / 32;
 
-  // Row-clamped operand base pointers (clamp => finite garbage that only
-  // reaches discarded outputs).
-  const int a0r = lm < B ? lm : B - 1;
-  const int a1r = (16 + lm) < B ? (16 + lm) : B - 1;
-  const int wr = g * H + (j0 + lm < H ? j0 + lm : H - 1);
-  const bf16* pa0 = h_prev + (int64_t)a0r * H;
-  const bf16* pa1 = h_prev + (int64_t)a1r * H;
-  const bf16* pw = W_h + (int64_t)wr * H;
-
-  const int full = H / 32;         // unguarded 32-k steps
-  const bool tail = (full * 32) < H;
+  const bf16x8* pa = reinterpret_cast<const bf16x8*>(h_pack) + l;
+  const bf16x8* pw = reinterpret_cast<const bf16x8*>(W_pack) +
+                     ((int64_t)blockIdx.x * 4 + g) * KS * 64 + l;
 
   f32x4 acc0 = {}, acc1 = {};
-
-  // 8-step chunks: 24 x 16-B loads issued per iteration before any wait,
-  // amortizing L2 latency over 16 MFMAs (hipcc groups loads at the top of
-  // the iteration; cross-iteration register rotation does not survive the
-  // scheduler, so work WITH that shape instead).
   int ks = 0;
-  for (; ks + 8 <= full; ks += 8) {
+  for (; ks + 8 <= KS; ks += 8) {
     bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      a0v[u] = *reinterpret_cast<const bf16x8*>(pa0 + (ks + u) * 32 + lk);
-      a1v[u] = *reinterpret_cast<const bf16x8*>(pa1 + (ks + u) * 32 + lk);
-      bwv[u] = *reinterpret_cast<const bf16x8*>(pw + (ks + u) * 32 + lk);
+      a0v[u] = pa[(ks + u) * 128];
+      a1v[u] = pa[(ks + u) * 128 + 64];
+      bwv[u] = pw[(ks + u) * 64];
     }
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
@@ -100,20 +143,12 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
       acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
     }
   }
-  for (; ks < full; ++ks) {
-    bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + ks * 32 + lk);
-    bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + ks * 32 + lk);
-    bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pw + ks * 32 + lk);
+  for (; ks < KS; ++ks) {
+    bf16x8 a0v = pa[ks * 128];
+    bf16x8 a1v = pa[ks * 128 + 64];
+    bf16x8 bwv = pw[ks * 64];
     acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
     acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
-  }
-  if (tail) {
-    const int k = full * 32 + lk;
-    bf16x8 a0t = frag_tail(pa0, k, H);
-    bf16x8 a1t = frag_tail(pa1, k, H);
-    bf16x8 bwt = frag_tail(pw, k, H);
-    acc0 = mfma_16x16x32_bf16(a0t, bwt, acc0);
-    acc1 = mfma_16x16x32_bf16(a1t, bwt, acc1);
   }
 
   // ---- exchange gate tiles through LDS -----------------------------------
@@ -126,7 +161,7 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
   }
   __syncthreads();
 
-  // ---- pointwise cell update ---------------------------------------------
+  // ---- pointwise cell update (dual h store: row-major + packed) ----------
   for (int idx = threadIdx.x; idx < B * 16; idx += CELL_THREADS) {
     const int b = idx / 16;
     const int jj = idx % 16;
@@ -143,8 +178,14 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
     float n_ = tanhf(gn);
     float c_ = f_ * c_prev[(int64_t)b * H + j] + i_ * n_;
     float h_ = o_ * tanhf(c_);
+    const bf16 hb = f2bf(h_);
     c_out[(int64_t)b * H + j] = c_;
-    h_out[(int64_t)b * H + j] = f2bf(h_);
+    h_out[(int64_t)b * H + j] = hb;
+    {  // packed slot for the next step's A operand
+      const int ks2 = j / 32, sub = j % 32;
+      const int pl = (b & 15) + 16 * (sub / 8);
+      h_pack_out[(((int64_t)ks2 * 2 + b / 16) * 64 + pl) * 8 + sub % 8] = hb;
+    }
     gates_out[gbase + 0 * H] = f2bf(i_);
     gates_out[gbase + 1 * H] = f2bf(f_);
     gates_out[gbase + 2 * H] = f2bf(o_);
@@ -152,27 +193,23 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
   }
 }
 
-void launch_lstm_cell_fwd(const bf16* h_prev, const float* c_prev,
-                          const bf16* gx, const bf16* W_h, bf16* h_out,
-                          float* c_out, bf16* gates_out, int B, int H,
-                          hipStream_t stream) {
+void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
+                          const bf16* gx, const bf16* W_pack, bf16* h_out,
+                          bf16* h_pack_out, float* c_out, bf16* gates_out,
+                          int B, int H, hipStream_t stream) {
   hipLaunchKernelGGL((lstm_cell_fwd_kernel<32>), dim3(cdiv(H, 16)),
-                     dim3(CELL_THREADS), 0, stream, h_prev, c_prev, gx, W_h,
-                     h_out, c_out, gates_out, B, H);
+                     dim3(CELL_THREADS), 0, stream, h_pack, c_prev, gx,
+                     W_pack, h_out, h_pack_out, c_out, gates_out, B, H);
 }
 
 // ---------------------------------------------------------------------------
-// Backward: per-timestep elementwise dgate kernel
+// Backward: per-timestep elementwise dgate kernel (dual dG store)
 // ---------------------------------------------------------------------------
-// dh_t = dy_t + dh_rec; tc = tanh(c_t); do = dh*tc
-// dct = dc + dh*o*(1-tc^2); di = dct*n; df = dct*c_prev; dn = dct*i
-// pre-activation grads via the sigmoid/tanh local derivatives; carried
-// dc <- dct * f.
 __global__ void lstm_cell_bwd_elt_kernel(
     const bf16* __restrict__ dy, const float* __restrict__ dh_rec,
     float* __restrict__ dc, const bf16* __restrict__ gates,
     const float* __restrict__ c_prev, const float* __restrict__ c_new,
-    bf16* __restrict__ dG, int B, int H) {
+    bf16* __restrict__ dG, bf16* __restrict__ dG_pack, int B, int H) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= B * H) return;
   int b = idx / H, j = idx % H;
@@ -188,71 +225,64 @@ __global__ void lstm_cell_bwd_elt_kernel(
   float di = dct * n_;
   float df = dct * c_prev[idx];
   float dn = dct * i_;
-  dG[gbase + 0 * H] = f2bf(di * i_ * (1.f - i_));
-  dG[gbase + 1 * H] = f2bf(df * f_ * (1.f - f_));
-  dG[gbase + 2 * H] = f2bf(do_ * o_ * (1.f - o_));
-  dG[gbase + 3 * H] = f2bf(dn * (1.f - n_ * n_));
+  const bf16 v[4] = {f2bf(di * i_ * (1.f - i_)), f2bf(df * f_ * (1.f - f_)),
+                     f2bf(do_ * o_ * (1.f - o_)), f2bf(dn * (1.f - n_ * n_))};
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    const int k = g * H + j;  // column in [B, 4H]
+    dG[gbase + (int64_t)g * H] = v[g];
+    const int ks = k / 32, sub = k % 32;
+    const int pl = (b & 15) + 16 * (sub / 8);
+    dG_pack[(((int64_t)ks * 2 + b / 16) * 64 + pl) * 8 + sub % 8] = v[g];
+  }
   dc[idx] = dct * f_;
 }
 
 void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
                               const bf16* gates, const float* c_prev,
-                              const float* c_new, bf16* dG, int B, int H,
-                              hipStream_t stream) {
+                              const float* c_new, bf16* dG, bf16* dG_pack,
+                              int B, int H, hipStream_t stream) {
   int n = B * H;
   hipLaunchKernelGGL(lstm_cell_bwd_elt_kernel, dim3(cdiv(n, 256)), dim3(256),
-                     0, stream, dy, dh_rec, dc, gates, c_prev, c_new, dG, B,
-                     H);
+                     0, stream, dy, dh_rec, dc, gates, c_prev, c_new, dG,
+                     dG_pack, B, H);
 }
 
 // ---------------------------------------------------------------------------
-// Skinny-M NT GEMM: C[M<=32, N] (fp32) = A[M,K] bf16 @ B[N,K]^T bf16
+// Skinny-M NT GEMM, packed operands: C[M<=32, N] f32 = A @ B^T
 // ---------------------------------------------------------------------------
-// The recurrent backward hop dh_rec = dG_t @ W_h (B = the W_h^T shadow,
-// [H, 4H] row-major, K contiguous). Grid = ceil(N/16); the 4 waves
-// round-robin the 32-wide K steps (wave w takes steps w, w+4, ...) so
-// every wave sees uniform full steps; one LDS reduction at the end.
-// Same streaming design as the forward cell: row-clamped direct global
-// fragment loads, depth-4 prefetch, element guards only on the K tail.
-// Requires K % 8 == 0 (K = 4H here).
+// The recurrent backward hop dh_rec = dG_t @ W_h. A_pack is the packed
+// dG ([KS][2][64][8]); W_pack is the packed W_h^T shadow
+// ([ceil(N/16)][KS][64][8]). Grid = ceil(N/16); the 4 waves round-robin
+// the K steps; one LDS reduction at the end. Fully uniform — pads are
+// zero in both packs.
 template <int MAXB>
-__global__ __launch_bounds__(CELL_THREADS) void smallm_gemm_nt_kernel(
-    const bf16* __restrict__ A,  // [M, K]
-    const bf16* __restrict__ B_, // [N, K]
-    float* __restrict__ C,       // [M, N]
-    int M, int N, int K) {
+__global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
+    const bf16* __restrict__ A_pack, const bf16* __restrict__ W_pack,
+    float* __restrict__ C, int M, int N, int K) {
   __shared__ float red[4 * MAXB * 16];
 
   const int n0 = blockIdx.x * 16;
   const int w = wave_id();
   const int l = lane_id();
   const int lm = l & 15;
-  const int lk = (l >> 4) * 8;
+  const int KS = (K + 31) / 32;
 
-  const int a0r = lm < M ? lm : M - 1;
-  const int a1r = (16 + lm) < M ? (16 + lm) : M - 1;
-  const int br = n0 + lm < N ? n0 + lm : N - 1;
-  const bf16* pa0 = A + (int64_t)a0r * K;
-  const bf16* pa1 = A + (int64_t)a1r * K;
-  const bf16* pb = B_ + (int64_t)br * K;
-
-  const int nsteps = (K + 31) / 32;
-  const int full = K / 32;
+  const bf16x8* pa = reinterpret_cast<const bf16x8*>(A_pack) + l;
+  const bf16x8* pw = reinterpret_cast<const bf16x8*>(W_pack) +
+                     (int64_t)blockIdx.x * KS * 64 + l;
 
   f32x4 acc0 = {}, acc1 = {};
-
-  // wave w owns steps w, w+4, w+8, ... ; i-th owned step = w + 4i.
-  // 8-owned-step chunks (24 loads in flight per iteration).
-  const int nown = (full - w + 3) / 4;  // owned FULL steps
+  const int nown = (KS - w + 3) / 4;  // steps owned by this wave (w + 4i)
   int i = 0;
   for (; i + 8 <= nown; i += 8) {
     bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const int kk = (w + 4 * (i + u)) * 32 + lk;
-      a0v[u] = *reinterpret_cast<const bf16x8*>(pa0 + kk);
-      a1v[u] = *reinterpret_cast<const bf16x8*>(pa1 + kk);
-      bwv[u] = *reinterpret_cast<const bf16x8*>(pb + kk);
+      const int ks = w + 4 * (i + u);
+      a0v[u] = pa[ks * 128];
+      a1v[u] = pa[ks * 128 + 64];
+      bwv[u] = pw[ks * 64];
     }
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
@@ -261,21 +291,12 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_gemm_nt_kernel(
     }
   }
   for (; i < nown; ++i) {
-    const int kk = (w + 4 * i) * 32 + lk;
-    bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + kk);
-    bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + kk);
-    bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pb + kk);
+    const int ks = w + 4 * i;
+    bf16x8 a0v = pa[ks * 128];
+    bf16x8 a1v = pa[ks * 128 + 64];
+    bf16x8 bwv = pw[ks * 64];
     acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
     acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
-  }
-  // K tail step (K % 32 != 0), owned by wave (full % 4)
-  if (full < nsteps && w == (full % 4)) {
-    const int k = full * 32 + lk;
-    bf16x8 a0t = frag_tail(pa0, k, K);
-    bf16x8 a1t = frag_tail(pa1, k, K);
-    bf16x8 bt = frag_tail(pb, k, K);
-    acc0 = mfma_16x16x32_bf16(a0t, bt, acc0);
-    acc1 = mfma_16x16x32_bf16(a1t, bt, acc1);
   }
 
   const int fr0 = (l >> 4) * 4;
@@ -294,10 +315,12 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_gemm_nt_kernel(
   }
 }
 
-void launch_smallm_gemm_nt(const bf16* A, const bf16* B, float* C, int M,
-                           int N, int K, hipStream_t stream) {
-  hipLaunchKernelGGL((smallm_gemm_nt_kernel<32>), dim3(cdiv(N, 16)),
-                     dim3(CELL_THREADS), 0, stream, A, B, C, M, N, K);
+void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream) {
+  hipLaunchKernelGGL((smallm_packed_nt_kernel<32>), dim3(cdiv(N, 16)),
+                     dim3(CELL_THREADS), 0, stream, A_pack, W_pack, C, M, N,
+                     K);
 }
 
 }  // namespace zamd

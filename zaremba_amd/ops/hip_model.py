@@ -24,6 +24,10 @@ from .. import _C
 from .hip_ops import DropoutFn, EmbeddingFn, LinearFn, LstmLayerFn
 
 
+def _ks(k):
+    return (k + 31) // 32
+
+
 class _LayerWorkspace:
     def __init__(self, T: int, B: int, H: int, device):
         bf, f32 = torch.bfloat16, torch.float32
@@ -36,6 +40,12 @@ class _LayerWorkspace:
         self.dG = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
         self.dh_rec = torch.zeros(B, H, dtype=f32, device=device)
         self.dc = torch.zeros(B, H, dtype=f32, device=device)
+        # fragment-packed workspaces (zero-prefilled: pad rows/K-tails
+        # must read as 0.0 in the packed cell kernels)
+        self.h_pack = torch.zeros(T + 1, _ks(H) * 2 * 64 * 8, dtype=bf,
+                                  device=device)
+        self.dG_pack = torch.zeros(_ks(4 * H) * 2 * 64 * 8, dtype=bf,
+                                   device=device)
 
 
 class _LayerRuntime:
@@ -44,12 +54,19 @@ class _LayerRuntime:
     def __init__(self, layer, device):
         self.H = layer.hidden_size
         self.Hin = layer.input_size
+        H = self.H
         bf = torch.bfloat16
         self.Wx = torch.empty_like(layer.W_x, dtype=bf, device=device)
         self.Wh = torch.empty_like(layer.W_h, dtype=bf, device=device)
-        self.WxT = torch.empty(self.Hin, 4 * self.H, dtype=bf, device=device)
-        self.WhT = torch.empty(self.H, 4 * self.H, dtype=bf, device=device)
-        self.bias_sum = torch.empty(4 * self.H, dtype=torch.float32,
+        self.WxT = torch.empty(self.Hin, 4 * H, dtype=bf, device=device)
+        self.WhT = torch.empty(H, 4 * H, dtype=bf, device=device)
+        # fragment-packed W_h (forward cell) and W_h^T (backward hop)
+        nb = (H + 15) // 16
+        self.WhP = torch.empty(nb * 4 * _ks(H) * 64 * 8, dtype=bf,
+                               device=device)
+        self.WhTP = torch.empty(nb * _ks(4 * H) * 64 * 8, dtype=bf,
+                                device=device)
+        self.bias_sum = torch.empty(4 * H, dtype=torch.float32,
                                     device=device)
         self.ws: Optional[_LayerWorkspace] = None
 
@@ -59,6 +76,8 @@ class _LayerRuntime:
         self.Wh.copy_(layer.W_h)
         e.transpose_bf16(self.Wx, self.WxT)
         e.transpose_bf16(self.Wh, self.WhT)
+        e.pack_gated_w(self.Wh, self.WhP, self.H, 4, self.H)
+        e.pack_gated_w(self.WhT, self.WhTP, self.H, 1, 4 * self.H)
         torch.add(layer.b_x, layer.b_h, out=self.bias_sum)
 
     def ensure_ws(self, T, B, device):
@@ -176,10 +195,12 @@ class HipModel:
             e.sgd_update(p.data.view(-1), p.grad.reshape(-1),
                          sh.view(-1) if sh is not None else None,
                          self.norm2, max_norm, lr, grad_scale)
-        # refresh the derived shadows (transposes + folded biases)
+        # refresh the derived shadows (transposes + packs + folded biases)
         for rt, layer in zip(self.layers, m.rnns):
             e.transpose_bf16(rt.Wx, rt.WxT)
             e.transpose_bf16(rt.Wh, rt.WhT)
+            e.pack_gated_w(rt.Wh, rt.WhP, rt.H, 4, rt.H)
+            e.pack_gated_w(rt.WhT, rt.WhTP, rt.H, 1, 4 * rt.H)
             torch.add(layer.b_x, layer.b_h, out=rt.bias_sum)
         e.transpose_bf16(self.fc.W, self.fc.WT)
         return (self.norm2.sqrt() * grad_scale).reshape(())

@@ -160,7 +160,8 @@ class LstmLayerFn(torch.autograd.Function):
                rt.bias_sum, False, False)
         rt.ws.h_all[0].copy_(h0.to(torch.bfloat16))
         rt.ws.c_all[0].copy_(c0.to(torch.float32))
-        e.lstm_seq_fwd(rt.ws.gx, rt.Wh, rt.ws.h_all, rt.ws.c_all, rt.ws.gates)
+        e.lstm_seq_fwd(rt.ws.gx, rt.WhP, rt.ws.h_all, rt.ws.h_pack,
+                       rt.ws.c_all, rt.ws.gates)
         ctx.save_for_backward(x2)
         ctx.rt = rt
         out = rt.ws.h_all[1:]          # [T, B, H] bf16 view (aliases ws)
@@ -179,8 +180,8 @@ class LstmLayerFn(torch.autograd.Function):
         Hin = x2.size(2)
         ws.dY.copy_(dY.to(torch.bfloat16))
         ws.dc.zero_()
-        e.lstm_seq_bwd(ws.dY, ws.gates, ws.c_all, rt.WhT, ws.dG, ws.dh_rec,
-                       ws.dc)
+        e.lstm_seq_bwd(ws.dY, ws.gates, ws.c_all, rt.WhTP, ws.dG, ws.dG_pack,
+                       ws.dh_rec, ws.dc)
         TB = T * B
         dG2 = ws.dG.view(TB, 4 * H)
         # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x —
@@ -197,7 +198,7 @@ class LstmLayerFn(torch.autograd.Function):
         e.gemm(dGT, xT, dWx, None, False, False)
         dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
         e.gemm(dG2, rt.WxT, dx, None, False, False)
-        db = torch.empty(4 * H, dtype=torch.float32, device=x2.device)
+        db = torch.zeros(4 * H, dtype=torch.float32, device=x2.device)
         e.colsum_bf16(dG2, db)  # grads of b_x and b_h are identical
         return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db.clone(),
                 None)
