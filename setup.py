@@ -20,6 +20,7 @@ ext = CUDAExtension(
         "zaremba_amd/csrc/ext_bind.hip",
         "zaremba_amd/csrc/gemm.hip",
         "zaremba_amd/csrc/lstm.hip",
+        "zaremba_amd/csrc/lstm_persistent.hip",
         "zaremba_amd/csrc/elementwise.hip",
     ],
     extra_compile_args={

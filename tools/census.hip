@@ -1,0 +1,108 @@
+// Residency census + grid-barrier probe for the persistent LSTM kernel
+// geometry (256 threads, ~141 KB dynamic LDS). Prints, for a range of
+// grid sizes, whether all blocks were co-resident (each arrives at a
+// counter and spins until it reads the full count, with a bounded spin).
+//
+// Build: hipcc --offload-arch=gfx950 -O2 tools/census.hip -o tools/census
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+typedef __attribute__((address_space(1))) unsigned int gu32;
+
+__global__ void census_kernel(unsigned int* cnt, unsigned int* result,
+                              int nb) {
+  extern __shared__ char smem[];
+  smem[threadIdx.x] = 0;  // touch LDS so it is really allocated
+  if (threadIdx.x == 0) {
+    gu32* c = (gu32*)(uintptr_t)cnt;
+    __hip_atomic_fetch_add(c, 1u, RLX_AGENT);
+    unsigned int spins = 0;
+    while (__hip_atomic_load(c, RLX_AGENT) < (unsigned int)nb) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > 3000000u) {
+        atomicAdd(result + 1, 1u);  // timed out
+        return;
+      }
+    }
+    atomicAdd(result, 1u);  // saw the full count
+  }
+}
+
+// the same two-level barrier as lstm_persistent.hip, iterated
+__device__ bool xcd_barrier(unsigned int* pstate, int grp, int nbg,
+                            int ngroups, unsigned int gen,
+                            unsigned int* fail) {
+  __shared__ int ok_s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    ok_s = 1;
+    gu32* st = (gu32*)(uintptr_t)pstate;
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    unsigned int t = __hip_atomic_fetch_add(&st[grp], 1u, RLX_AGENT);
+    if (t == gen * nbg - 1) {
+      unsigned int tt = __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
+      if (tt == gen * ngroups - 1) {
+        for (int x = 0; x < 8; ++x)
+          __hip_atomic_store(&st[9 + x], gen, RLX_AGENT);
+      }
+    }
+    unsigned int spins = 0;
+    while (__hip_atomic_load(&st[9 + grp], RLX_AGENT) < gen) {
+      __builtin_amdgcn_s_sleep(8);
+      if (++spins > 3000000u) {
+        atomicAdd(fail, 1u);
+        ok_s = 0;
+        break;
+      }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  return ok_s != 0;
+}
+
+__global__ void barrier_probe_kernel(unsigned int* pstate,
+                                     unsigned int* fail, unsigned int* done,
+                                     int iters, int nb) {
+  extern __shared__ char smem[];
+  smem[threadIdx.x] = 0;
+  const int grp = blockIdx.x & 7;
+  const int ngroups = nb < 8 ? nb : 8;
+  const int nbg = (nb - grp + 7) / 8;
+  for (int t = 1; t <= iters; ++t) {
+    if (!xcd_barrier(pstate, grp, nbg, ngroups, (unsigned int)t, fail))
+      return;
+  }
+  if (threadIdx.x == 0) atomicAdd(done, 1u);
+}
+
+int main() {
+  size_t lds = (size_t)(4 * 6 + 20) * 1512 * 2 + 4 * 32 * 16 * 4;
+  printf("LDS request: %zu bytes\n", lds);
+  unsigned int* buf;
+  (void)hipMalloc(&buf, 4096);
+  for (int nb : {128, 200, 240, 248, 250, 252, 256, 260}) {
+    (void)hipMemset(buf, 0, 4096);
+    hipLaunchKernelGGL(census_kernel, dim3(nb), dim3(256), lds, 0, buf,
+                       buf + 64, nb);
+    (void)hipDeviceSynchronize();
+    unsigned int res[2];
+    (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
+    printf("census nb=%3d: full=%u timeout=%u\n", nb, res[0], res[1]);
+  }
+  // barrier probe at the largest resident size
+  for (int nb : {250, 248, 240}) {
+    (void)hipMemset(buf, 0, 4096);
+    hipLaunchKernelGGL(barrier_probe_kernel, dim3(nb), dim3(256), lds, 0,
+                       buf, buf + 64, buf + 65, 200, nb);
+    (void)hipDeviceSynchronize();
+    unsigned int res[2];
+    (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
+    printf("barrier nb=%3d iters=200: fail=%u done=%u\n", nb, res[0],
+           res[1]);
+  }
+  return 0;
+}

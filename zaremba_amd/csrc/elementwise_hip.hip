@@ -346,20 +346,25 @@ void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
                      dst, R, C);
 }
 
-// column sum of a bf16 matrix -> f32 (bias grads: db = colsum(dG))
+// column sum of a bf16 matrix -> f32 (bias grads: db = colsum(dG)).
+// 2-D grid: row-chunks in y accumulate via atomicAdd (out pre-zeroed by
+// the wrapper); a single serial column walk was latency-bound (165 us).
 __global__ void colsum_bf16_kernel(const bf16* __restrict__ in,
                                    float* __restrict__ out, int Rr, int Cc) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= Cc) return;
+  int r0 = blockIdx.y * 64;
+  int r1 = min(r0 + 64, Rr);
   float acc = 0.f;
-  for (int r = 0; r < Rr; ++r) acc += bf2f(in[(int64_t)r * Cc + c]);
-  out[c] = acc;
+  for (int r = r0; r < r1; ++r) acc += bf2f(in[(int64_t)r * Cc + c]);
+  atomicAdd(out + c, acc);
 }
 
 void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                         hipStream_t stream) {
-  hipLaunchKernelGGL(colsum_bf16_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
-                     stream, in, out, R, C);
+  dim3 grid(cdiv(C, 256), cdiv(R, 64));
+  hipLaunchKernelGGL(colsum_bf16_kernel, grid, dim3(256), 0, stream, in, out,
+                     R, C);
 }
 
 }  // namespace zamd

@@ -99,11 +99,28 @@ struct GraphCache {
 };
 static GraphCache g_fwd_graphs, g_bwd_graphs;
 static bool g_use_graphs = true;
+static bool g_use_persistent = true;
 
-static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_pack,
-                              bf16* h_all, bf16* h_pack, float* c_all,
-                              bf16* gates, int T, int B, int H,
+// The persistent forward needs every block co-resident and one cell
+// element per thread: B*HS <= 256, H even, B <= 32.
+static bool persistent_ok(int B, int H) {
+  return g_use_persistent && B <= 32 && (H % 2) == 0 &&
+         B * persistent_hs(H) <= 256;
+}
+
+static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,
+                              const bf16* W_pack, bf16* h_all, bf16* h_pack,
+                              float* c_all, bf16* gates,
+                              unsigned long long* hgran,  // barrier state
+                              unsigned int* abort_flag, int T, int B, int H,
                               hipStream_t stream) {
+  if (persistent_ok(B, H)) {
+    HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
+    launch_lstm_persistent_fwd(gx, W_h, h_all, c_all, gates,
+                               reinterpret_cast<unsigned int*>(hgran),
+                               abort_flag, T, B, H, stream);
+    return;
+  }
   const int64_t hstep = (int64_t)B * H;
   const int64_t gstep = (int64_t)B * 4 * H;
   const int64_t pstep = (int64_t)((H + 31) / 32) * 2 * 64 * 8;
@@ -117,32 +134,43 @@ static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_pack,
 }
 
 // h_all/c_all are [T+1, B, H] with slot 0 pre-filled with (h0, c0);
-// h_pack is the zero-prefilled [T+1, KS*2*64*8] packed-h workspace.
-static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_pack,
+// h_pack is the zero-prefilled [T+1, KS*2*64*8] packed-h workspace;
+// hgran the [2*B*H/2] u64 granule buffer; abort a u32 flag.
+static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
+                         const torch::Tensor& W_pack,
                          torch::Tensor& h_all, torch::Tensor& h_pack,
-                         torch::Tensor& c_all, torch::Tensor& gates) {
+                         torch::Tensor& c_all, torch::Tensor& gates,
+                         torch::Tensor& hgran, torch::Tensor& abort_flag) {
   int T = gx.size(0), B = gx.size(1);
   int H = h_all.size(2);
   TORCH_CHECK(gx.size(2) == 4 * H, "gx must be [T,B,4H]");
   const bf16* gxp = bf_ptr(gx);
+  const bf16* whraw = bf_ptr(W_h);
   const bf16* whp = bf_ptr(W_pack);
   bf16* hp = bf_ptr_mut(h_all);
   bf16* hpk = bf_ptr_mut(h_pack);
   float* cp = f_ptr_mut(c_all);
   bf16* gp = bf_ptr_mut(gates);
+  auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
+  auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
   auto stream = current_stream();
-  if (!g_use_graphs) {
-    lstm_seq_fwd_body(gxp, whp, hp, hpk, cp, gp, T, B, H, stream);
+  // The persistent path is one memset + one kernel: no graph needed, and
+  // replaying a grid-synchronized persistent kernel from a graph hangs
+  // intermittently on ROCm 7.x — always run it eagerly.
+  if (!g_use_graphs || persistent_ok(B, H)) {
+    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, hg, ab, T, B, H,
+                      stream);
     return;
   }
-  std::vector<uintptr_t> key{(uintptr_t)gxp, (uintptr_t)whp, (uintptr_t)hp,
-                             (uintptr_t)hpk, (uintptr_t)cp, (uintptr_t)gp,
-                             (uintptr_t)T, (uintptr_t)B, (uintptr_t)H};
+  std::vector<uintptr_t> key{(uintptr_t)gxp, (uintptr_t)whraw, (uintptr_t)whp,
+                             (uintptr_t)hp, (uintptr_t)hpk, (uintptr_t)cp,
+                             (uintptr_t)gp, (uintptr_t)hg, (uintptr_t)T,
+                             (uintptr_t)B, (uintptr_t)H};
   auto it = g_fwd_graphs.cache.find(key);
   if (it == g_fwd_graphs.cache.end()) {
     hipStream_t cs = g_fwd_graphs.capture_stream();
     HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
-    lstm_seq_fwd_body(gxp, whp, hp, hpk, cp, gp, T, B, H, cs);
+    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, hg, ab, T, B, H, cs);
     hipGraph_t graph;
     HIP_CHECK(hipStreamEndCapture(cs, &graph));
     hipGraphExec_t exec;
@@ -212,6 +240,7 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
 }
 
 static void set_use_graphs(bool v) { g_use_graphs = v; }
+static void set_use_persistent(bool v) { g_use_persistent = v; }
 static void clear_graphs() {
   for (auto& kv : g_fwd_graphs.cache) hipGraphExecDestroy(kv.second);
   for (auto& kv : g_bwd_graphs.cache) hipGraphExecDestroy(kv.second);
@@ -365,5 +394,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose_bf16", &zamd::transpose_bf16);
   m.def("colsum_bf16", &zamd::colsum_bf16);
   m.def("set_use_graphs", &zamd::set_use_graphs);
+  m.def("set_use_persistent", &zamd::set_use_persistent);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

@@ -33,6 +33,14 @@ void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
                              float* C, int M, int N, int K,
                              hipStream_t stream);
 
+// lstm_persistent.hip — one launch for a whole layer unroll
+int persistent_hs(int H);
+void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
+                                float* c_all, bf16* gates_out,
+                                unsigned int* pstate,
+                                unsigned int* abort_flag, int T, int B,
+                                int H, hipStream_t stream);
+
 // elementwise.hip
 void launch_embedding_fwd(const bf16* W, const int64_t* idx, bf16* out,
                           int N, int H, hipStream_t stream);

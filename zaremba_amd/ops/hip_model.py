@@ -46,6 +46,10 @@ class _LayerWorkspace:
                                   device=device)
         self.dG_pack = torch.zeros(_ks(4 * H) * 2 * 64 * 8, dtype=bf,
                                    device=device)
+        # persistent-forward barrier state (17 words, re-zeroed per call
+        # by the sequence driver) + abort flag
+        self.hgran = torch.zeros(32, dtype=torch.int64, device=device)
+        self.abort = torch.zeros(1, dtype=torch.int32, device=device)
 
 
 class _LayerRuntime:
@@ -106,6 +110,11 @@ class HipModel:
         if self.device.type != "cuda":
             raise RuntimeError("HipModel requires a ROCm GPU device")
         self.e = _C.ext()
+        import os
+        if os.environ.get("ZAREMBA_AMD_GRAPHS", "1") == "0":
+            self.e.set_use_graphs(False)
+        if os.environ.get("ZAREMBA_AMD_PERSISTENT", "1") == "0":
+            self.e.set_use_persistent(False)
         self.compute_dtype = torch.bfloat16
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,

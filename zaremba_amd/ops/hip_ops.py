@@ -160,8 +160,8 @@ class LstmLayerFn(torch.autograd.Function):
                rt.bias_sum, False, False)
         rt.ws.h_all[0].copy_(h0.to(torch.bfloat16))
         rt.ws.c_all[0].copy_(c0.to(torch.float32))
-        e.lstm_seq_fwd(rt.ws.gx, rt.WhP, rt.ws.h_all, rt.ws.h_pack,
-                       rt.ws.c_all, rt.ws.gates)
+        e.lstm_seq_fwd(rt.ws.gx, rt.Wh, rt.WhP, rt.ws.h_all, rt.ws.h_pack,
+                       rt.ws.c_all, rt.ws.gates, rt.ws.hgran, rt.ws.abort)
         ctx.save_for_backward(x2)
         ctx.rt = rt
         out = rt.ws.h_all[1:]          # [T, B, H] bf16 view (aliases ws)
