@@ -134,6 +134,8 @@ def test_lstm_cell_fwd_step(ext, B, H):
 def _run_seq_fwd(ext, T, B, H, seed=5):
     torch.manual_seed(seed)
     KS = (H + 31) // 32
+    hs = ext.persistent_hs(H)
+    nb = (H + hs - 1) // hs
     gx = (torch.randn(T, B, 4 * H, device=dev()) * 0.5).to(torch.bfloat16)
     W_h = (torch.randn(4 * H, H, device=dev()) * 0.02).to(torch.bfloat16)
     WhP = torch.empty(((H + 15) // 16) * 4 * KS * 64 * 8, device=dev(),
@@ -143,43 +145,51 @@ def _run_seq_fwd(ext, T, B, H, seed=5):
     h_pack = torch.zeros(T + 1, KS * 2 * 64 * 8, device=dev(),
                          dtype=torch.bfloat16)
     c_all = torch.zeros(T + 1, B, H, device=dev(), dtype=torch.float32)
-    gates = torch.empty(T, B, 4 * H, device=dev(), dtype=torch.bfloat16)
-    hgran = torch.zeros(B * H, device=dev(), dtype=torch.int64)
+    gates = torch.zeros(T, B, 4 * H, device=dev(), dtype=torch.bfloat16)
+    rec = torch.zeros(T * nb * B * 6 * hs, device=dev(),
+                      dtype=torch.bfloat16)
+    hgran = torch.zeros(32, device=dev(), dtype=torch.int64)
     abort = torch.zeros(1, device=dev(), dtype=torch.int32)
     h_all[0] = (torch.randn(B, H, device=dev()) * 0.3).to(torch.bfloat16)
     c_all[0] = torch.randn(B, H, device=dev()) * 0.3
-    ext.lstm_seq_fwd(gx, W_h, WhP, h_all, h_pack, c_all, gates, hgran, abort)
+    c0 = c_all[0].clone()
+    ext.lstm_seq_fwd(gx, W_h, WhP, h_all, h_pack, c_all, gates, rec, hgran,
+                     abort)
     assert abort.item() == 0, "persistent forward aborted (spin timeout)"
-    return gx, W_h, h_all, c_all, gates
+    return gx, W_h, h_all, c_all, c0
 
 
 @pytest.mark.parametrize("H", [650, 1500])
 def test_lstm_seq_fwd_matches_step_loop(ext, H):
+    """h trajectory + final c vs the fp32 oracle (the persistent kernel
+    carries c in registers and writes back only the final state)."""
     T, B = 6, 20
-    gx, W_h, h_all, c_all, _ = _run_seq_fwd(ext, T, B, H)
+    gx, W_h, h_all, c_all, c0 = _run_seq_fwd(ext, T, B, H)
     h = h_all[0]
-    c = c_all[0].clone()
+    c = c0
     for t in range(T):
         h_ref, c_ref, _ = _ref_cell(h, c, gx[t], W_h)
         assert rel_err(h_all[t + 1], h_ref) < 3e-2, t
         h = h_all[t + 1]  # carry the kernel's bf16 h to isolate per-step err
-        c = c_all[t + 1].clone()
+        c = c_ref
+    assert rel_err(c_all[T], c) < 3e-2
 
 
 def test_lstm_seq_fwd_persistent_matches_per_step(ext):
     """The persistent one-launch path and the per-step fallback must agree
-    bitwise (same math, same fragment order)."""
+    on the h trajectory and carried state (same math, same MFMA order)."""
     T, B, H = 7, 20, 1500
-    gx1, _, h1, c1, g1 = _run_seq_fwd(ext, T, B, H, seed=11)
+    gx1, _, h1, c1, _ = _run_seq_fwd(ext, T, B, H, seed=11)
+    h1 = h1.clone()
+    cT1 = c1[T].clone()
     ext.set_use_persistent(False)
     try:
-        gx2, _, h2, c2, g2 = _run_seq_fwd(ext, T, B, H, seed=11)
+        gx2, _, h2, c2, _ = _run_seq_fwd(ext, T, B, H, seed=11)
     finally:
         ext.set_use_persistent(True)
     assert torch.equal(gx1, gx2)
     assert torch.equal(h1, h2)
-    assert torch.equal(g1, g2)
-    assert torch.allclose(c1, c2, atol=1e-6)
+    assert torch.allclose(cT1, c2[T], atol=1e-6)
 
 
 def test_lstm_layer_autograd_matches_eager(ext):

@@ -100,6 +100,7 @@ struct GraphCache {
 static GraphCache g_fwd_graphs, g_bwd_graphs;
 static bool g_use_graphs = true;
 static bool g_use_persistent = true;
+static bool g_use_persistent_bwd = false;  // measured slower; kept for A/B
 
 // The persistent forward needs every block co-resident and one cell
 // element per thread: B*HS <= 256, H even, B <= 32.
@@ -110,13 +111,13 @@ static bool persistent_ok(int B, int H) {
 
 static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,
                               const bf16* W_pack, bf16* h_all, bf16* h_pack,
-                              float* c_all, bf16* gates,
+                              float* c_all, bf16* gates, bf16* rec,
                               unsigned long long* hgran,  // barrier state
                               unsigned int* abort_flag, int T, int B, int H,
                               hipStream_t stream) {
   if (persistent_ok(B, H)) {
     HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
-    launch_lstm_persistent_fwd(gx, W_h, h_all, c_all, gates,
+    launch_lstm_persistent_fwd(gx, W_h, h_all, c_all, rec,
                                reinterpret_cast<unsigned int*>(hgran),
                                abort_flag, T, B, H, stream);
     return;
@@ -124,12 +125,15 @@ static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,
   const int64_t hstep = (int64_t)B * H;
   const int64_t gstep = (int64_t)B * 4 * H;
   const int64_t pstep = (int64_t)((H + 31) / 32) * 2 * 64 * 8;
+  const int HSp = persistent_hs(H);
+  const int64_t rstep = (int64_t)((H + HSp - 1) / HSp) * B * 6 * HSp;
   launch_pack_a(h_all, h_pack, B, H, stream);  // slot 0 = h0
   for (int t = 0; t < T; ++t) {
     launch_lstm_cell_fwd(h_pack + t * pstep, c_all + t * hstep,
                          gx + t * gstep, W_pack, h_all + (t + 1) * hstep,
                          h_pack + (t + 1) * pstep, c_all + (t + 1) * hstep,
-                         gates + t * gstep, B, H, stream);
+                         gates + t * gstep, rec + t * rstep, B, H, HSp,
+                         stream);
   }
 }
 
@@ -140,7 +144,8 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
                          const torch::Tensor& W_pack,
                          torch::Tensor& h_all, torch::Tensor& h_pack,
                          torch::Tensor& c_all, torch::Tensor& gates,
-                         torch::Tensor& hgran, torch::Tensor& abort_flag) {
+                         torch::Tensor& rec, torch::Tensor& hgran,
+                         torch::Tensor& abort_flag) {
   int T = gx.size(0), B = gx.size(1);
   int H = h_all.size(2);
   TORCH_CHECK(gx.size(2) == 4 * H, "gx must be [T,B,4H]");
@@ -151,6 +156,7 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
   bf16* hpk = bf_ptr_mut(h_pack);
   float* cp = f_ptr_mut(c_all);
   bf16* gp = bf_ptr_mut(gates);
+  bf16* rp = bf_ptr_mut(rec);
   auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
   auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
   auto stream = current_stream();
@@ -158,7 +164,7 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
   // replaying a grid-synchronized persistent kernel from a graph hangs
   // intermittently on ROCm 7.x — always run it eagerly.
   if (!g_use_graphs || persistent_ok(B, H)) {
-    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, hg, ab, T, B, H,
+    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, rp, hg, ab, T, B, H,
                       stream);
     return;
   }
@@ -170,7 +176,8 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
   if (it == g_fwd_graphs.cache.end()) {
     hipStream_t cs = g_fwd_graphs.capture_stream();
     HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
-    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, hg, ab, T, B, H, cs);
+    lstm_seq_fwd_body(gxp, whraw, whp, hp, hpk, cp, gp, rp, hg, ab, T, B, H,
+                      cs);
     hipGraph_t graph;
     HIP_CHECK(hipStreamEndCapture(cs, &graph));
     hipGraphExec_t exec;
@@ -182,18 +189,28 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
 }
 
 static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
-                              const float* c_all, const bf16* WT_pack,
+                              const bf16* rec, const float* c_all,
+                              const bf16* W_h_T, const bf16* WT_pack,
                               bf16* dG, bf16* dG_pack, float* dh_rec,
-                              float* dc, int T, int B, int H,
+                              float* dc, unsigned long long* hgran,
+                              unsigned int* abort_flag, int T, int B, int H,
                               hipStream_t stream) {
+  if (g_use_persistent_bwd && persistent_ok(B, H)) {
+    HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
+    launch_lstm_persistent_bwd(dY, rec, W_h_T, dG,
+                               reinterpret_cast<unsigned int*>(hgran),
+                               abort_flag, T, B, H, stream);
+    return;
+  }
   const int64_t hstep = (int64_t)B * H;
   const int64_t gstep = (int64_t)B * 4 * H;
+  const int HSp = persistent_hs(H);
+  const int64_t rstep = (int64_t)((H + HSp - 1) / HSp) * B * 6 * HSp;
   for (int t = T - 1; t >= 0; --t) {
     launch_lstm_cell_bwd_elt(dY + t * hstep,
                              (t == T - 1) ? nullptr : dh_rec, dc,
-                             gates + t * gstep, c_all + t * hstep,
-                             c_all + (t + 1) * hstep, dG + t * gstep, dG_pack,
-                             B, H, stream);
+                             rec + t * rstep, dG + t * gstep, dG_pack,
+                             B, H, HSp, stream);
     launch_smallm_packed_nt(dG_pack, WT_pack, dh_rec, B, H, 4 * H, stream);
   }
 }
@@ -202,25 +219,32 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
 // dG_pack the zero-prefilled packed-dG workspace; dh_rec a [B,H] f32
 // workspace. WT_pack is the packed W_h^T shadow.
 static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
-                         const torch::Tensor& c_all,
+                         const torch::Tensor& rec, const torch::Tensor& c_all,
+                         const torch::Tensor& W_h_T,
                          const torch::Tensor& WT_pack, torch::Tensor& dG,
                          torch::Tensor& dG_pack, torch::Tensor& dh_rec,
-                         torch::Tensor& dc) {
+                         torch::Tensor& dc, torch::Tensor& hgran,
+                         torch::Tensor& abort_flag) {
   int T = dY.size(0), B = dY.size(1), H = dY.size(2);
   const bf16* dyp = bf_ptr(dY);
   const bf16* gp = bf_ptr(gates);
+  const bf16* rp = bf_ptr(rec);
   const float* cp = f_ptr(c_all);
+  const bf16* whtp = bf_ptr(W_h_T);
   const bf16* wtp = bf_ptr(WT_pack);
   bf16* dgp = bf_ptr_mut(dG);
   bf16* dgpk = bf_ptr_mut(dG_pack);
   float* dhp = f_ptr_mut(dh_rec);
   float* dcp = f_ptr_mut(dc);
+  auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
+  auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
   auto stream = current_stream();
-  if (!g_use_graphs) {
-    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dgpk, dhp, dcp, T, B, H, stream);
+  if (!g_use_graphs || (g_use_persistent_bwd && persistent_ok(B, H))) {
+    lstm_seq_bwd_body(dyp, gp, rp, cp, whtp, wtp, dgp, dgpk, dhp, dcp, hg,
+                      ab, T, B, H, stream);
     return;
   }
-  std::vector<uintptr_t> key{(uintptr_t)dyp, (uintptr_t)gp, (uintptr_t)cp,
+  std::vector<uintptr_t> key{(uintptr_t)dyp, (uintptr_t)gp, (uintptr_t)rp,
                              (uintptr_t)wtp, (uintptr_t)dgp, (uintptr_t)dgpk,
                              (uintptr_t)dhp, (uintptr_t)dcp, (uintptr_t)T,
                              (uintptr_t)B, (uintptr_t)H};
@@ -228,7 +252,8 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
   if (it == g_bwd_graphs.cache.end()) {
     hipStream_t cs = g_bwd_graphs.capture_stream();
     HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
-    lstm_seq_bwd_body(dyp, gp, cp, wtp, dgp, dgpk, dhp, dcp, T, B, H, cs);
+    lstm_seq_bwd_body(dyp, gp, rp, cp, whtp, wtp, dgp, dgpk, dhp, dcp, hg,
+                      ab, T, B, H, cs);
     hipGraph_t graph;
     HIP_CHECK(hipStreamEndCapture(cs, &graph));
     hipGraphExec_t exec;
@@ -241,6 +266,7 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
 
 static void set_use_graphs(bool v) { g_use_graphs = v; }
 static void set_use_persistent(bool v) { g_use_persistent = v; }
+static void set_use_persistent_bwd(bool v) { g_use_persistent_bwd = v; }
 static void clear_graphs() {
   for (auto& kv : g_fwd_graphs.cache) hipGraphExecDestroy(kv.second);
   for (auto& kv : g_bwd_graphs.cache) hipGraphExecDestroy(kv.second);
@@ -282,9 +308,13 @@ static void lstm_cell_fwd_step(const torch::Tensor& h_prev,
   auto hp = pack_a_tmp(h_prev);
   auto wp = pack_w_tmp(W_h, H, 4, H);
   auto hpo = torch::zeros_like(hp);
+  const int HSp = persistent_hs(H);
+  const int nb = (H + HSp - 1) / HSp;
+  auto rec = torch::zeros({(int64_t)nb * B * 6 * HSp}, h_prev.options());
   launch_lstm_cell_fwd(bf_ptr(hp), f_ptr(c_prev), bf_ptr(gx), bf_ptr(wp),
                        bf_ptr_mut(h_out), bf_ptr_mut(hpo), f_ptr_mut(c_out),
-                       bf_ptr_mut(gates), B, H, current_stream());
+                       bf_ptr_mut(gates), bf_ptr_mut(rec), B, H, HSp,
+                       current_stream());
 }
 
 static void smallm_gemm_nt(const torch::Tensor& A, const torch::Tensor& B,
@@ -395,5 +425,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_bf16", &zamd::colsum_bf16);
   m.def("set_use_graphs", &zamd::set_use_graphs);
   m.def("set_use_persistent", &zamd::set_use_persistent);
+  m.def("persistent_hs", &zamd::persistent_hs);
+  m.def("set_use_persistent_bwd", &zamd::set_use_persistent_bwd);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

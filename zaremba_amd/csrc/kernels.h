@@ -24,11 +24,11 @@ void launch_pack_a(const bf16* A, bf16* out, int B, int K,
 void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
                           const bf16* gx, const bf16* W_pack, bf16* h_out,
                           bf16* h_pack_out, float* c_out, bf16* gates_out,
-                          int B, int H, hipStream_t stream);
+                          bf16* rec, int B, int H, int HSp,
+                          hipStream_t stream);
 void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
-                              const bf16* gates, const float* c_prev,
-                              const float* c_new, bf16* dG, bf16* dG_pack,
-                              int B, int H, hipStream_t stream);
+                              const bf16* rec, bf16* dG, bf16* dG_pack,
+                              int B, int H, int HSp, hipStream_t stream);
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
                              float* C, int M, int N, int K,
                              hipStream_t stream);
@@ -36,7 +36,12 @@ void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
 // lstm_persistent.hip — one launch for a whole layer unroll
 int persistent_hs(int H);
 void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
-                                float* c_all, bf16* gates_out,
+                                float* c_all, bf16* rec,
+                                unsigned int* pstate,
+                                unsigned int* abort_flag, int T, int B,
+                                int H, hipStream_t stream);
+void launch_lstm_persistent_bwd(const bf16* dY, const bf16* rec,
+                                const bf16* W_h_T, bf16* dG,
                                 unsigned int* pstate,
                                 unsigned int* abort_flag, int T, int B,
                                 int H, hipStream_t stream);

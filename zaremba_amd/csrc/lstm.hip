@@ -113,7 +113,9 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
     bf16* __restrict__ h_pack_out,     // packed h_{t+1}
     float* __restrict__ c_out,         // [B, H]
     bf16* __restrict__ gates_out,      // [B, 4H] post-activation i,f,o,n
-    int B, int H) {
+    bf16* __restrict__ rec,            // [NB][B][6][HSp] block record for
+                                       // this t (persistent_hs layout)
+    int B, int H, int HSp) {
   __shared__ float gbuf[4 * MAXB * 16];
 
   const int j0 = blockIdx.x * 16;  // hidden-unit slice
@@ -175,11 +177,23 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
     float f_ = 1.f / (1.f + __expf(-gf));
     float o_ = 1.f / (1.f + __expf(-go));
     float n_ = tanhf(gn);
-    float c_ = f_ * c_prev[(int64_t)b * H + j] + i_ * n_;
-    float h_ = o_ * tanhf(c_);
+    const float cp = c_prev[(int64_t)b * H + j];
+    float c_ = f_ * cp + i_ * n_;
+    const float tc = tanhf(c_);
+    float h_ = o_ * tc;
     const bf16 hb = f2bf(h_);
     c_out[(int64_t)b * H + j] = c_;
     h_out[(int64_t)b * H + j] = hb;
+    {  // block record in the persistent layout (backward reads this)
+      const int blk = j / HSp, jr = j % HSp;
+      bf16* r = rec + (((int64_t)blk * B + b) * 6) * HSp;
+      r[0 * HSp + jr] = f2bf(i_);
+      r[1 * HSp + jr] = f2bf(f_);
+      r[2 * HSp + jr] = f2bf(o_);
+      r[3 * HSp + jr] = f2bf(n_);
+      r[4 * HSp + jr] = f2bf(tc);
+      r[5 * HSp + jr] = f2bf(cp);
+    }
     {  // packed slot for the next step's A operand
       const int ks2 = j / 32, sub = j % 32;
       const int pl = (b & 15) + 16 * (sub / 8);
@@ -195,10 +209,12 @@ __global__ __launch_bounds__(CELL_THREADS) void lstm_cell_fwd_kernel(
 void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
                           const bf16* gx, const bf16* W_pack, bf16* h_out,
                           bf16* h_pack_out, float* c_out, bf16* gates_out,
-                          int B, int H, hipStream_t stream) {
+                          bf16* rec, int B, int H, int HSp,
+                          hipStream_t stream) {
   hipLaunchKernelGGL((lstm_cell_fwd_kernel<32>), dim3(cdiv(H, 16)),
                      dim3(CELL_THREADS), 0, stream, h_pack, c_prev, gx,
-                     W_pack, h_out, h_pack_out, c_out, gates_out, B, H);
+                     W_pack, h_out, h_pack_out, c_out, gates_out, rec, B, H,
+                     HSp);
 }
 
 // ---------------------------------------------------------------------------
@@ -206,23 +222,26 @@ void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
 // ---------------------------------------------------------------------------
 __global__ void lstm_cell_bwd_elt_kernel(
     const bf16* __restrict__ dy, const float* __restrict__ dh_rec,
-    float* __restrict__ dc, const bf16* __restrict__ gates,
-    const float* __restrict__ c_prev, const float* __restrict__ c_new,
-    bf16* __restrict__ dG, bf16* __restrict__ dG_pack, int B, int H) {
+    float* __restrict__ dc, const bf16* __restrict__ rec,
+    bf16* __restrict__ dG, bf16* __restrict__ dG_pack, int B, int H,
+    int HSp) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= B * H) return;
   int b = idx / H, j = idx % H;
   const int64_t gbase = (int64_t)b * 4 * H + j;
-  float i_ = bf2f(gates[gbase + 0 * H]);
-  float f_ = bf2f(gates[gbase + 1 * H]);
-  float o_ = bf2f(gates[gbase + 2 * H]);
-  float n_ = bf2f(gates[gbase + 3 * H]);
+  const int blk = j / HSp, jr = j % HSp;
+  const bf16* r = rec + (((int64_t)blk * B + b) * 6) * HSp;
+  float i_ = bf2f(r[0 * HSp + jr]);
+  float f_ = bf2f(r[1 * HSp + jr]);
+  float o_ = bf2f(r[2 * HSp + jr]);
+  float n_ = bf2f(r[3 * HSp + jr]);
+  float tc = bf2f(r[4 * HSp + jr]);
+  float cprev = bf2f(r[5 * HSp + jr]);
   float dh = bf2f(dy[idx]) + (dh_rec ? dh_rec[idx] : 0.f);
-  float tc = tanhf(c_new[idx]);
   float do_ = dh * tc;
   float dct = dc[idx] + dh * o_ * (1.f - tc * tc);
   float di = dct * n_;
-  float df = dct * c_prev[idx];
+  float df = dct * cprev;
   float dn = dct * i_;
   const bf16 v[4] = {f2bf(di * i_ * (1.f - i_)), f2bf(df * f_ * (1.f - f_)),
                      f2bf(do_ * o_ * (1.f - o_)), f2bf(dn * (1.f - n_ * n_))};
@@ -238,13 +257,11 @@ __global__ void lstm_cell_bwd_elt_kernel(
 }
 
 void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
-                              const bf16* gates, const float* c_prev,
-                              const float* c_new, bf16* dG, bf16* dG_pack,
-                              int B, int H, hipStream_t stream) {
+                              const bf16* rec, bf16* dG, bf16* dG_pack,
+                              int B, int H, int HSp, hipStream_t stream) {
   int n = B * H;
   hipLaunchKernelGGL(lstm_cell_bwd_elt_kernel, dim3(cdiv(n, 256)), dim3(256),
-                     0, stream, dy, dh_rec, dc, gates, c_prev, c_new, dG,
-                     dG_pack, B, H);
+                     0, stream, dy, dh_rec, dc, rec, dG, dG_pack, B, H, HSp);
 }
 
 // ---------------------------------------------------------------------------

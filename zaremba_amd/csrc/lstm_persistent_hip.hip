@@ -1,33 +1,39 @@
 #include "hip/hip_runtime.h"
-// Persistent LSTM forward kernel for gfx950 — one launch runs the whole
-// T-step unroll of a layer.
+// Persistent LSTM kernels for gfx950 — one launch runs a whole layer
+// unroll (forward or backward).
 //
 // Why (measured, see profiles/): with one launch per timestep, the W_h
 // weight (18 MB bf16 at H=1500) misses L2 on EVERY launch — kernel
 // boundaries do not retain it across the 8 per-XCD L2s — so each of the
 // 70 cell launches of a training step re-pulls the full weight from
 // LLC/HBM and runs ~90% wave-wait. Here each of the NB resident
-// workgroups holds its W_h slice in LDS for the entire unroll and the
-// only per-step global traffic is the h broadcast (60 KB, L2-amplified).
+// workgroups holds its weight slice in LDS for the entire unroll; the
+// per-step global traffic is just the h (fwd) / dgates (bwd) broadcast.
 //
 // Inter-step exchange (guide §6 Guideline 16, placement-independent):
-//   * h_all[t+1] itself is the exchange buffer — every slot is written
-//     exactly once inside the launch, so there is no reader-side reuse
-//     of addresses and no ring/parity logic. (A first version broadcast
-//     h as 8-byte tagged granules; relaxed agent loads are memory-side
-//     served, so every consumer re-pulled the full payload from the
-//     fabric — 30 MB/step — and it measured slower than relaunching.)
-//   * between steps: every block does {plain h stores -> __syncthreads ->
-//     lane-0 agent release fence + vmcnt drain -> arrival} into an
-//     XCD-grouped two-level counter barrier (monotonic counters, epoch
-//     generations — no per-step state reset), then one agent acquire.
-//     Spins are bounded; on timeout the block sets *abort and exits.
-//   * the barrier state words (8 group counters, 1 top counter, 8
-//     generation words) must be zeroed before every launch (the driver
-//     issues the hipMemsetAsync).
+//   * the exchanged tensor's own [t] slot is the buffer — every slot is
+//     written exactly once inside a launch, so no ring/parity logic and
+//     no reader-side address reuse. (A granule-tagged variant measured
+//     slower: relaxed agent loads are memory-side served, so every
+//     consumer re-pulled the payload from the fabric.)
+//   * producers store the exchanged values WRITE-THROUGH (relaxed
+//     agent-scope = sc1) and every wave drains vmcnt before arriving at
+//     an XCD-grouped two-level counter barrier (monotonic counters,
+//     epoch generations); consumers take one agent acquire after the
+//     generation flip, then read PLAIN (L2-amplified). Spins are
+//     bounded; on timeout a block sets *abort and exits.
+//   * barrier state (8 group counters, top counter, 8 generations) is
+//     zeroed before every launch by the sequence driver.
 //
-// Compute per step is the same 4-wave MFMA gate reduction + pointwise
-// cell update as the per-step cell kernel (bitwise-identical results).
+// Scattered-store avoidance (measured: 2-byte stores into [B,4H]-layout
+// gates cost ~90 MB/launch of read-modify-write line fills): everything
+// the backward needs per (t, block) is written CONTIGUOUSLY as a
+// 6-channel block record rec[t][blk][b][6][HS] = {i, f, o, n,
+// tanh(c_t+1), c_t(bf16)}. c itself stays fp32 in registers across the
+// unroll (exact state); only the backward's df term sees bf16 c_prev.
+// Standard-layout c_all/h_all are written where the framework needs
+// them (h every step — it IS the layer output and the exchange buffer —
+// c only at the final step, the carried state).
 #include "common.h"
 
 namespace zamd {
@@ -35,6 +41,7 @@ namespace zamd {
 typedef __attribute__((address_space(1))) unsigned int gu32;
 
 constexpr int PCELL_THREADS = 256;
+constexpr int REC_CH = 6;  // i,f,o,n,tanh_c_new,c_prev channels
 
 // HS must be even; NB = ceil(H/HS) <= 250 so the grid is co-resident.
 int persistent_hs(int H) {
@@ -44,12 +51,12 @@ int persistent_hs(int H) {
 #define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
 
 // pstate layout: [0..7] group arrival counters, [8] top counter,
-// [9..16] group generation words. Monotonic: gen = step index (1-based).
+// [9..16] group generation words. Monotonic: gen = 1-based step index.
 DEV_INLINE bool xcd_grid_barrier(unsigned int* pstate, int grp, int nbg,
                                  int ngroups, unsigned int gen,
                                  unsigned int* abort_flag) {
   __shared__ int ok_s;
-  // every wave drains its own write-through h stores before arriving
+  // every wave drains its own write-through stores before arriving
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   if (threadIdx.x == 0) {
@@ -66,7 +73,7 @@ DEV_INLINE bool xcd_grid_barrier(unsigned int* pstate, int grp, int nbg,
     unsigned int spins = 0;
     while (__hip_atomic_load(&st[9 + grp], RLX_AGENT) < gen) {
       __builtin_amdgcn_s_sleep(8);
-      if (++spins > 50000000u) {
+      if (++spins > 20000000u) {
         atomicOr(abort_flag, 1u);
         ok_s = 0;
         break;
@@ -78,14 +85,26 @@ DEV_INLINE bool xcd_grid_barrier(unsigned int* pstate, int grp, int nbg,
   return ok_s != 0;
 }
 
+// paired write-through store of two adjacent bf16 values
+DEV_INLINE void store_pair_wt(bf16* p, bf16 lo, bf16 hi) {
+  unsigned int packed = (unsigned int)__builtin_bit_cast(unsigned short, lo) |
+                        ((unsigned int)__builtin_bit_cast(unsigned short, hi)
+                         << 16);
+  __hip_atomic_store((gu32*)(uintptr_t)p, packed, RLX_AGENT);
+}
+
+// ===========================================================================
+// Forward
+// ===========================================================================
 template <int MAXB>
 __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     const bf16* __restrict__ gx,     // [T, B, 4H]
     const bf16* __restrict__ W_h,    // [4H, H] bf16 shadow
     bf16* __restrict__ h_all,        // [T+1, B, H]; slot 0 = h0 (input)
-    float* __restrict__ c_all,       // [T+1, B, H]; slot 0 = c0 (input)
-    bf16* __restrict__ gates_out,    // [T, B, 4H]
-    unsigned int* __restrict__ pstate,  // 17 zeroed words (barrier state)
+    float* __restrict__ c_all,       // [T+1, B, H]; slot 0 = c0 (input);
+                                     // only slot T is written back
+    bf16* __restrict__ rec,          // [T][NB][B][6][HS] block records
+    unsigned int* __restrict__ pstate,
     unsigned int* __restrict__ abort_flag,
     int T, int B, int H, int HS) {
   const int KS = (H + 31) / 32;
@@ -99,7 +118,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
   const int NB = (H + HS - 1) / HS;
   const int grp = blockIdx.x & 7;
   const int ngroups = NB < 8 ? NB : 8;
-  const int nbg = (NB - grp + 7) / 8;  // blocks in this group
+  const int nbg = (NB - grp + 7) / 8;
 
   const int j0 = blockIdx.x * HS;
   const int g = wave_id();
@@ -125,14 +144,12 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     }
     *reinterpret_cast<bf16x8*>(Ws + (int64_t)row * KPAD + kv) = v;
   }
-  // zero the h image K-tail once (never rewritten)
   for (int idx = t_; idx < B * (KPAD - H); idx += PCELL_THREADS) {
     const int b = idx / (KPAD - H);
     const int k = H + idx % (KPAD - H);
     hs[(int64_t)b * KPAD + k] = (bf16)0.f;
   }
 
-  // ---- per-thread cell state (thread t_ owns element (b, jj)) ------------
   const int own_b = t_ / HS;
   const int own_jj = t_ % HS;
   const bool own = (t_ < B * HS) && (j0 + own_jj < H);
@@ -147,14 +164,13 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
 
   for (int t = 0; t < T; ++t) {
     if (t > 0) {
-      // all blocks' h_all[t] stores done + visible before anyone reads
       if (!xcd_grid_barrier(pstate, grp, nbg, ngroups, (unsigned int)t,
                             abort_flag))
         return;
     } else {
       __syncthreads();
     }
-    // ---- stage h_t into the LDS image (plain global loads) ---------------
+    // ---- stage h_t into the LDS image (plain loads, L2-amplified) --------
     {
       const bf16* hsrc = h_all + (int64_t)t * B * H;
       const int vecs = (H + 7) / 8;
@@ -174,7 +190,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- gate MFMA reduction (wave g -> gate g, [32 x 16(HS)] tile) ------
+    // ---- gate MFMA reduction (wave g -> gate g) --------------------------
     f32x4 acc0 = {}, acc1 = {};
     {
       const bf16* pw = Ws + (int64_t)(g * HS + wc) * KPAD;
@@ -212,7 +228,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- pointwise cell update + stores ----------------------------------
+    // ---- pointwise cell update ------------------------------------------
     if (own) {
       const int b = own_b, jj = own_jj;
       const int j = j0 + jj;
@@ -225,20 +241,23 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       float f_ = 1.f / (1.f + __expf(-gf));
       float o_ = 1.f / (1.f + __expf(-go));
       float n_ = tanhf(gn);
+      const float c_prev = c_reg;
       c_reg = f_ * c_reg + i_ * n_;
-      const float h_ = o_ * tanhf(c_reg);
-      const int64_t hoff = ((int64_t)(t + 1) * B + b) * H + j;
-      c_all[hoff] = c_reg;
-      gates_out[gxbase + 0 * H] = f2bf(i_);
-      gates_out[gxbase + 1 * H] = f2bf(f_);
-      gates_out[gxbase + 2 * H] = f2bf(o_);
-      gates_out[gxbase + 3 * H] = f2bf(n_);
+      const float tc = tanhf(c_reg);
+      const float h_ = o_ * tc;
+      // contiguous block record (i,f,o,n,tanh_c_new,c_prev)
+      bf16* r = rec + ((((int64_t)t * NB + blockIdx.x) * B + b) * REC_CH) * HS;
+      r[0 * HS + jj] = f2bf(i_);
+      r[1 * HS + jj] = f2bf(f_);
+      r[2 * HS + jj] = f2bf(o_);
+      r[3 * HS + jj] = f2bf(n_);
+      r[4 * HS + jj] = f2bf(tc);
+      r[5 * HS + jj] = f2bf(c_prev);
       hbuf[b * HS + jj] = f2bf(h_);
+      if (t == T - 1) c_all[((int64_t)T * B + b) * H + j] = c_reg;
     }
     __syncthreads();
-    // publish h_{t+1} as paired write-through (sc1) stores — the only
-    // data other blocks read inside this launch. No release fence needed
-    // (R1 write-through form); each wave drains vmcnt at the barrier.
+    // ---- publish h_{t+1}: paired write-through stores --------------------
     {
       bf16* hdst = h_all + (int64_t)(t + 1) * B * H;
       for (int i = t_; i < B * HS / 2; i += PCELL_THREADS) {
@@ -246,13 +265,8 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
         const int jj = (i % (HS / 2)) * 2;
         const int j = j0 + jj;
         if (j < H) {
-          const bf16 h0v = hbuf[b * HS + jj];
-          const bf16 h1v = (j + 1 < H) ? hbuf[b * HS + jj + 1] : (bf16)0.f;
-          unsigned int packed =
-              (unsigned int)__builtin_bit_cast(unsigned short, h0v) |
-              ((unsigned int)__builtin_bit_cast(unsigned short, h1v) << 16);
-          gu32* p = (gu32*)(uintptr_t)(hdst + (int64_t)b * H + j);
-          __hip_atomic_store(p, packed, RLX_AGENT);
+          store_pair_wt(hdst + (int64_t)b * H + j, hbuf[b * HS + jj],
+                        (j + 1 < H) ? hbuf[b * HS + jj + 1] : (bf16)0.f);
         }
       }
     }
@@ -260,7 +274,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
 }
 
 void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
-                                float* c_all, bf16* gates_out,
+                                float* c_all, bf16* rec,
                                 unsigned int* pstate,
                                 unsigned int* abort_flag, int T, int B,
                                 int H, hipStream_t stream) {
@@ -272,7 +286,203 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
                (size_t)B * HS * 2 + 16;
   hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32>), dim3(NB),
                      dim3(PCELL_THREADS), lds, stream, gx, W_h, h_all, c_all,
-                     gates_out, pstate, abort_flag, T, B, H, HS);
+                     rec, pstate, abort_flag, T, B, H, HS);
+}
+
+// ===========================================================================
+// Backward
+// ===========================================================================
+// Per step t = T-1..0, each block (owning h-units [j0, j0+HS)):
+//   1. elementwise dgate math for its units from the block record +
+//      dY[t] + its register-carried (dc, dh_rec) state,
+//   2. publish its dG[t] columns (write-through pairs into the standard
+//      [T,B,4H] dG tensor — consumed in-launch AND later by the dW
+//      GEMMs),
+//   3. grid barrier,
+//   4. recurrent hop: dh_rec = dG[t] @ W_h for its HS output columns —
+//      W_h^T slice LDS-resident, dG staged chunk-wise (plain loads),
+//      4 waves split K, LDS reduction back to per-thread dh_rec.
+constexpr int BWD_CHUNK = 1024;  // K elements staged per chunk
+
+template <int MAXB>
+__global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_bwd_kernel(
+    const bf16* __restrict__ dY,     // [T, B, H]
+    const bf16* __restrict__ rec,    // [T][NB][B][6][HS]
+    const bf16* __restrict__ W_h_T,  // [H, 4H] transposed shadow
+    bf16* __restrict__ dG,           // [T, B, 4H] out (write-through)
+    unsigned int* __restrict__ pstate,
+    unsigned int* __restrict__ abort_flag,
+    int T, int B, int H, int HS) {
+  const int K = 4 * H;
+  const int CPAD = BWD_CHUNK + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* Ws = reinterpret_cast<bf16*>(smem);                // [HS][KWPAD]
+  // rows cover whole staged chunks (reads go to chunk-rounded k); the
+  // zero fill past K keeps 0*garbage out of the MFMA
+  const int KWPAD = ((K + BWD_CHUNK - 1) / BWD_CHUNK) * BWD_CHUNK + 8;
+  bf16* as = Ws + (int64_t)HS * KWPAD;                     // [B][CPAD]
+  float* gbuf = reinterpret_cast<float*>(as + (int64_t)B * CPAD);
+  bf16* dgbuf = reinterpret_cast<bf16*>(gbuf + 4 * MAXB * 16);  // [B][4][HS]
+
+  const int NB = (H + HS - 1) / HS;
+  const int grp = blockIdx.x & 7;
+  const int ngroups = NB < 8 ? NB : 8;
+  const int nbg = (NB - grp + 7) / 8;
+
+  const int j0 = blockIdx.x * HS;
+  const int w = wave_id();
+  const int l = lane_id();
+  const int lm = l & 15;
+  const int t_ = threadIdx.x;
+
+  // ---- load the W_h^T slice (rows j0..j0+HS of [H, 4H]) into LDS ---------
+  for (int idx = t_; idx < HS * (KWPAD / 8); idx += PCELL_THREADS) {
+    const int kv = (idx % (KWPAD / 8)) * 8;
+    const int row = idx / (KWPAD / 8);
+    bf16x8 v = {};
+    const int col = j0 + row;
+    if (col < H) {
+      const bf16* p = W_h_T + (int64_t)col * K + kv;
+      if (kv + 8 <= K) {
+        v = *reinterpret_cast<const bf16x8*>(p);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = (kv + e < K) ? p[e] : (bf16)0.f;
+      }
+    }
+    *reinterpret_cast<bf16x8*>(Ws + (int64_t)row * KWPAD + kv) = v;
+  }
+  // zero the chunk-image K-tail rows once (tail beyond staged chunk len)
+  for (int idx = t_; idx < B * (CPAD - BWD_CHUNK); idx += PCELL_THREADS) {
+    const int b = idx / (CPAD - BWD_CHUNK);
+    as[(int64_t)b * CPAD + BWD_CHUNK + idx % (CPAD - BWD_CHUNK)] = (bf16)0.f;
+  }
+
+  const int own_b = t_ / HS;
+  const int own_jj = t_ % HS;
+  const bool own = (t_ < B * HS) && (j0 + own_jj < H);
+  float dc_reg = 0.f, dh_rec = 0.f;
+
+  const int a0r = lm < B ? lm : B - 1;
+  const int a1r = (16 + lm) < B ? (16 + lm) : B - 1;
+  const int wc = (lm < HS ? lm : HS - 1);
+  const int fr0 = (l >> 4) * 4;
+  const int lk = (l >> 4) * 8;
+
+  for (int t = T - 1; t >= 0; --t) {
+    // ---- 1. elementwise dgates for own units -----------------------------
+    __syncthreads();  // previous iteration's dgbuf/as consumers done
+    if (own) {
+      const int b = own_b, jj = own_jj;
+      const int j = j0 + jj;
+      const bf16* r =
+          rec + ((((int64_t)t * NB + blockIdx.x) * B + b) * REC_CH) * HS;
+      const float i_ = bf2f(r[0 * HS + jj]);
+      const float f_ = bf2f(r[1 * HS + jj]);
+      const float o_ = bf2f(r[2 * HS + jj]);
+      const float n_ = bf2f(r[3 * HS + jj]);
+      const float tc = bf2f(r[4 * HS + jj]);
+      const float cprev = bf2f(r[5 * HS + jj]);
+      const float dh = bf2f(dY[((int64_t)t * B + b) * H + j]) + dh_rec;
+      const float do_ = dh * tc;
+      const float dct = dc_reg + dh * o_ * (1.f - tc * tc);
+      dgbuf[(b * 4 + 0) * HS + jj] = f2bf(dct * n_ * i_ * (1.f - i_));
+      dgbuf[(b * 4 + 1) * HS + jj] = f2bf(dct * cprev * f_ * (1.f - f_));
+      dgbuf[(b * 4 + 2) * HS + jj] = f2bf(do_ * o_ * (1.f - o_));
+      dgbuf[(b * 4 + 3) * HS + jj] = f2bf(dct * i_ * (1.f - n_ * n_));
+      dc_reg = dct * f_;
+    }
+    __syncthreads();
+    // ---- 2. publish dG[t] columns (paired write-through) -----------------
+    {
+      bf16* dst = dG + (int64_t)t * B * K;
+      for (int i = t_; i < B * 4 * HS / 2; i += PCELL_THREADS) {
+        const int b = i / (4 * HS / 2);
+        const int rem = i % (4 * HS / 2);
+        const int gg = rem / (HS / 2);
+        const int jj = (rem % (HS / 2)) * 2;
+        const int j = j0 + jj;
+        if (j < H) {
+          store_pair_wt(dst + (int64_t)b * K + gg * H + j,
+                        dgbuf[(b * 4 + gg) * HS + jj],
+                        (j + 1 < H) ? dgbuf[(b * 4 + gg) * HS + jj + 1]
+                                    : (bf16)0.f);
+        }
+      }
+    }
+    // ---- 3. grid barrier -------------------------------------------------
+    if (!xcd_grid_barrier(pstate, grp, nbg, ngroups, (unsigned int)(T - t),
+                          abort_flag))
+      return;
+    // ---- 4. recurrent hop: dh_rec = dG[t] @ W_h (own columns) ------------
+    f32x4 acc0 = {}, acc1 = {};
+    const bf16* dsrc = dG + (int64_t)t * B * K;
+    for (int k0 = 0; k0 < K; k0 += BWD_CHUNK) {
+      const int klen = min(BWD_CHUNK, K - k0);
+      // stage the chunk (plain loads; zero-fill tail inside the chunk)
+      for (int idx = t_; idx < B * (BWD_CHUNK / 8); idx += PCELL_THREADS) {
+        const int b = idx / (BWD_CHUNK / 8);
+        const int k = (idx % (BWD_CHUNK / 8)) * 8;
+        bf16x8 v = {};
+        if (k < klen) {
+          const bf16* p = dsrc + (int64_t)b * K + k0 + k;
+          if (k + 8 <= klen) {
+            v = *reinterpret_cast<const bf16x8*>(p);
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              v[e] = (k + e < klen) ? p[e] : (bf16)0.f;
+          }
+        }
+        *reinterpret_cast<bf16x8*>(as + (int64_t)b * CPAD + k) = v;
+      }
+      __syncthreads();
+      // waves split the chunk: wave w owns k in [w*256, w*256+256)
+      const bf16* pw = Ws + (int64_t)wc * KWPAD + k0;
+      const bf16* pa0 = as + (int64_t)a0r * CPAD;
+      const bf16* pa1 = as + (int64_t)a1r * CPAD;
+      const int kw0 = w * (BWD_CHUNK / 4);
+#pragma unroll
+      for (int ks = 0; ks < BWD_CHUNK / 4; ks += 32) {
+        const int k = kw0 + ks + lk;
+        bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + k);
+        bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + k);
+        bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pw + k);
+        acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
+        acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
+      }
+      __syncthreads();
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      gbuf[(w * MAXB + fr0 + r) * 16 + lm] = acc0[r];
+      gbuf[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+    }
+    __syncthreads();
+    if (own) {
+      const int b = own_b, jj = own_jj;
+      dh_rec = gbuf[(0 * MAXB + b) * 16 + jj] +
+               gbuf[(1 * MAXB + b) * 16 + jj] +
+               gbuf[(2 * MAXB + b) * 16 + jj] +
+               gbuf[(3 * MAXB + b) * 16 + jj];
+    }
+  }
+}
+
+void launch_lstm_persistent_bwd(const bf16* dY, const bf16* rec,
+                                const bf16* W_h_T, bf16* dG,
+                                unsigned int* pstate,
+                                unsigned int* abort_flag, int T, int B,
+                                int H, hipStream_t stream) {
+  const int HS = persistent_hs(H);
+  const int NB = cdiv(H, HS);
+  const int K = 4 * H;
+  const int KWPAD = ((K + BWD_CHUNK - 1) / BWD_CHUNK) * BWD_CHUNK + 8;
+  size_t lds = (size_t)HS * KWPAD * 2 + (size_t)B * (BWD_CHUNK + 8) * 2 +
+               4 * 32 * 16 * sizeof(float) + (size_t)B * 4 * HS * 2 + 16;
+  hipLaunchKernelGGL((lstm_persistent_bwd_kernel<32>), dim3(NB),
+                     dim3(PCELL_THREADS), lds, stream, dY, rec, W_h_T, dG,
+                     pstate, abort_flag, T, B, H, HS);
 }
 
 }  // namespace zamd

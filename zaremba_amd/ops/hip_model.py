@@ -46,8 +46,11 @@ class _LayerWorkspace:
                                   device=device)
         self.dG_pack = torch.zeros(_ks(4 * H) * 2 * 64 * 8, dtype=bf,
                                    device=device)
-        # persistent-forward barrier state (17 words, re-zeroed per call
-        # by the sequence driver) + abort flag
+        # persistent-kernel state: block records [T][NB][B][6][HS],
+        # barrier words (re-zeroed per call by the driver) + abort flag
+        hs = 2 * ((H + 499) // 500)
+        nb = (H + hs - 1) // hs
+        self.rec = torch.zeros(T * nb * B * 6 * hs, dtype=bf, device=device)
         self.hgran = torch.zeros(32, dtype=torch.int64, device=device)
         self.abort = torch.zeros(1, dtype=torch.int32, device=device)
 

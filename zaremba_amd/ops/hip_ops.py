@@ -161,7 +161,8 @@ class LstmLayerFn(torch.autograd.Function):
         rt.ws.h_all[0].copy_(h0.to(torch.bfloat16))
         rt.ws.c_all[0].copy_(c0.to(torch.float32))
         e.lstm_seq_fwd(rt.ws.gx, rt.Wh, rt.WhP, rt.ws.h_all, rt.ws.h_pack,
-                       rt.ws.c_all, rt.ws.gates, rt.ws.hgran, rt.ws.abort)
+                       rt.ws.c_all, rt.ws.gates, rt.ws.rec, rt.ws.hgran,
+                       rt.ws.abort)
         ctx.save_for_backward(x2)
         ctx.rt = rt
         out = rt.ws.h_all[1:]          # [T, B, H] bf16 view (aliases ws)
@@ -180,8 +181,9 @@ class LstmLayerFn(torch.autograd.Function):
         Hin = x2.size(2)
         ws.dY.copy_(dY.to(torch.bfloat16))
         ws.dc.zero_()
-        e.lstm_seq_bwd(ws.dY, ws.gates, ws.c_all, rt.WhTP, ws.dG, ws.dG_pack,
-                       ws.dh_rec, ws.dc)
+        e.lstm_seq_bwd(ws.dY, ws.gates, ws.rec, ws.c_all, rt.WhT, rt.WhTP,
+                       ws.dG, ws.dG_pack, ws.dh_rec, ws.dc, ws.hgran,
+                       ws.abort)
         TB = T * B
         dG2 = ws.dG.view(TB, 4 * H)
         # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x —
