@@ -32,6 +32,37 @@ DEV_INLINE int swz(int row, int byte_col) {
   return row * (BK * 2) + (byte_col ^ ((row & 7) << 4));
 }
 
+// ---- async global->LDS staging (interior fast path) -----------------------
+// One glds wave-instruction moves 1 KB: 8 rows x 128 B, lane l -> LDS byte
+// (wave-uniform base) + l*16. The LDS image stays linear; the XOR swizzle
+// moves to the per-lane SOURCE address and the matching XOR on the read
+// side (guide §5.4 rule 21: same involution on source and read).
+DEV_INLINE void glds16(const bf16* gsrc, bf16* lds_dst_uniform) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)gsrc,
+      (__attribute__((address_space(3))) void*)lds_dst_uniform, 16, 0, 0);
+}
+
+// Stage a [TILE][BK] k-contiguous tile via glds: TILE*BK*2/1024
+// instructions split across 4 waves.
+template <int TILE>
+DEV_INLINE void stage_glds_kc(const bf16* __restrict__ src, int ld, int row0,
+                              int k0, bf16* lds) {
+  constexpr int NINST = TILE * BK * 2 / 1024;     // 16 (TILE=128) or 4
+  constexpr int PER_WAVE = NINST / 4;
+  const int w = wave_id();
+  const int l = lane_id();
+#pragma unroll
+  for (int i = 0; i < PER_WAVE; ++i) {
+    const int inst = w * PER_WAVE + i;
+    const int row = inst * 8 + (l >> 3);          // 8 rows per instruction
+    const int colb = (l & 7) * 16;                // byte column 0..112
+    const int src_colb = colb ^ ((row & 7) << 4); // pre-swizzled source
+    glds16(src + (int64_t)(row0 + row) * ld + k0 + src_colb / 2,
+           lds + (int64_t)inst * 512);            // 1024 B = 512 bf16
+  }
+}
+
 // ---- staging: k-contiguous operand (NT layout), [BROWS][BK] tile ---------
 // Register-load phase: each thread grabs BROWS*BK/(256*8) bf16x8 vectors.
 template <int BROWS, bool GUARD>
@@ -136,8 +167,11 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     int K, int lda, int ldb, int ldc) {
   constexpr int WT = TILE / 2;       // wave tile (64 or 32)
   constexpr int NF = WT / 16;        // fragments per wave dim (4 or 2)
-  __shared__ bf16 As[TILE * BK];
-  __shared__ bf16 Bs[TILE * BK];
+  // one LDS object only (a second __shared__ forces vmcnt(0) before every
+  // ds_read when a glds is in flight — guide §5 trap (a)); double-buffered
+  __shared__ bf16 smem_all[4 * TILE * BK];
+  bf16* As = smem_all;               // [2][TILE*BK] A buffers
+  bf16* Bs = smem_all + 2 * TILE * BK;
 
   const int nbn = (N + TILE - 1) / TILE;
   const int bm = blockIdx.x / nbn;
@@ -157,7 +191,7 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
   bf16x8 va[TILE / 32], vb[TILE / 32];
 
   const int nk = (K + BK - 1) / BK;
-  const int k_full = K / BK;         // tiles with no K guard
+  const int k_full = K / BK;         // tiles with no K/MN guard
 
   auto load_tile = [&](int kt, bool guard_mn) {
     const bool gk = (kt >= k_full);
@@ -182,39 +216,70 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     }
   };
 
-  // The k-contiguous path software-pipelines (T14: hold tile kt+1 in
-  // registers across the MFMA phase). The transposed path stages
-  // synchronously — its 64 scalar-loaded values per thread would push the
-  // kernel to 278 registers (1 wave/SIMD) if held across the MFMA phase.
-  constexpr bool ASYNC = !(TRANS_A || TRANS_B);
-  if (ASYNC) load_tile(0, !interior);
-  for (int kt = 0; kt < nk; ++kt) {
-    if (!ASYNC) load_tile(kt, !interior);
-    __syncthreads();  // LDS consumers of tile kt-1 done
-    if (TRANS_A)
-      stage_write_tr<TILE>(As, va);
-    else
-      stage_write_kc<TILE>(As, va);
-    if (TRANS_B)
-      stage_write_tr<TILE>(Bs, vb);
-    else
-      stage_write_kc<TILE>(Bs, vb);
-    if (ASYNC && kt + 1 < nk) load_tile(kt + 1, !interior);  // issue early
-    __syncthreads();
+  auto mfma_phase = [&](const bf16* Ab, const bf16* Bb) {
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       bf16x8 af[NF], bfr[NF];
 #pragma unroll
       for (int i = 0; i < NF; ++i)
-        af[i] = frag_read<TILE>(As, wm + i * 16 + lm, kk + lk);
+        af[i] = frag_read<TILE>(Ab, wm + i * 16 + lm, kk + lk);
 #pragma unroll
       for (int j = 0; j < NF; ++j)
-        bfr[j] = frag_read<TILE>(Bs, wn + j * 16 + lm, kk + lk);
+        bfr[j] = frag_read<TILE>(Bb, wn + j * 16 + lm, kk + lk);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < NF; ++i)
 #pragma unroll
         for (int j = 0; j < NF; ++j)
           acc[i][j] = mfma_16x16x32_bf16(af[i], bfr[j], acc[i][j]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  if (!TRANS_A && !TRANS_B && interior && k_full > 0) {
+    // glds 2-phase pipeline (guide §5.5 T3 minimum form): stage tile t+1
+    // while computing tile t; __syncthreads() drains the in-flight glds.
+    stage_glds_kc<TILE>(A, lda, m0, 0, As);
+    stage_glds_kc<TILE>(B, ldb, n0, 0, Bs);
+    __syncthreads();
+    int cur = 0;
+    for (int kt = 0; kt < k_full; ++kt) {
+      if (kt + 1 < k_full) {
+        stage_glds_kc<TILE>(A, lda, m0, (kt + 1) * BK,
+                            As + (cur ^ 1) * TILE * BK);
+        stage_glds_kc<TILE>(B, ldb, n0, (kt + 1) * BK,
+                            Bs + (cur ^ 1) * TILE * BK);
+      }
+      mfma_phase(As + cur * TILE * BK, Bs + cur * TILE * BK);
+      __syncthreads();
+      cur ^= 1;
+    }
+    if (k_full < nk) {  // K tail: register-staged, guarded
+      stage_load_kc<TILE, true>(A, lda, m0, TILE, k_full * BK, K, va);
+      stage_load_kc<TILE, true>(B, ldb, n0, TILE, k_full * BK, K, vb);
+      stage_write_kc<TILE>(As + cur * TILE * BK, va);
+      stage_write_kc<TILE>(Bs + cur * TILE * BK, vb);
+      __syncthreads();
+      mfma_phase(As + cur * TILE * BK, Bs + cur * TILE * BK);
+    }
+  } else {
+    // register-staged path (edge blocks / transposed operands)
+    constexpr bool ASYNC = !(TRANS_A || TRANS_B);
+    if (ASYNC) load_tile(0, !interior);
+    for (int kt = 0; kt < nk; ++kt) {
+      if (!ASYNC) load_tile(kt, !interior);
+      __syncthreads();  // LDS consumers of tile kt-1 done
+      if (TRANS_A)
+        stage_write_tr<TILE>(As, va);
+      else
+        stage_write_kc<TILE>(As, va);
+      if (TRANS_B)
+        stage_write_tr<TILE>(Bs, vb);
+      else
+        stage_write_kc<TILE>(Bs, vb);
+      if (ASYNC && kt + 1 < nk) load_tile(kt + 1, !interior);  // issue early
+      __syncthreads();
+      mfma_phase(As, Bs);
     }
   }
 
