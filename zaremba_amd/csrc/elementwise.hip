@@ -247,6 +247,75 @@ void launch_norm2_accum(const float* g, int64_t n, float* accum,
                      n, accum);
 }
 
+// ---- multi-tensor variants (one launch for all params) --------------------
+// desc: per chunk of <=CHUNK_ELEMS elements, int64 fields. norm2:
+// {grad_ptr, n}; sgd: {master_ptr, grad_ptr, shadow_ptr|0, n}.
+constexpr int CHUNK_ELEMS = 65536;
+
+__global__ void norm2_mt_kernel(const int64_t* __restrict__ desc,
+                                float* __restrict__ accum) {
+  __shared__ float scratch[8];
+  const float* g = reinterpret_cast<const float*>(desc[blockIdx.x * 2]);
+  const int n = (int)desc[blockIdx.x * 2 + 1];
+  float acc = 0.f;
+  const int n4 = n / 4;
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  for (int i = threadIdx.x; i < n4; i += blockDim.x) {
+    float4 v = g4[i];
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+  }
+  if (threadIdx.x < (n & 3)) {
+    float v = g[n4 * 4 + threadIdx.x];
+    acc += v * v;
+  }
+  acc = block_reduce(acc, scratch, 0.f,
+                     [] __device__(float a, float b) { return a + b; });
+  if (threadIdx.x == 0) atomicAdd(accum, acc);
+}
+
+void launch_norm2_mt(const int64_t* desc, int nchunk, float* accum,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(norm2_mt_kernel, dim3(nchunk), dim3(256), 0, stream,
+                     desc, accum);
+}
+
+__global__ void sgd_mt_kernel(const int64_t* __restrict__ desc,
+                              const float* __restrict__ norm2,
+                              float max_norm, float lr, float grad_scale) {
+  float norm = sqrtf(*norm2) * grad_scale;
+  float coef = max_norm / (norm + 1e-6f);
+  coef = fminf(coef, 1.f) * lr * grad_scale;
+  float* m = reinterpret_cast<float*>(desc[blockIdx.x * 4]);
+  const float* g = reinterpret_cast<const float*>(desc[blockIdx.x * 4 + 1]);
+  bf16* sh = reinterpret_cast<bf16*>(desc[blockIdx.x * 4 + 2]);
+  const int n = (int)desc[blockIdx.x * 4 + 3];
+  const int n4 = n / 4;
+  float4* m4 = reinterpret_cast<float4*>(m);
+  const float4* g4 = reinterpret_cast<const float4*>(g);
+  bf16x4* s4 = reinterpret_cast<bf16x4*>(sh);
+  for (int i = threadIdx.x; i < n4; i += blockDim.x) {
+    float4 mv = m4[i];
+    float4 gv = g4[i];
+    mv.x -= coef * gv.x; mv.y -= coef * gv.y;
+    mv.z -= coef * gv.z; mv.w -= coef * gv.w;
+    m4[i] = mv;
+    if (sh) s4[i] = bf16x4{f2bf(mv.x), f2bf(mv.y), f2bf(mv.z), f2bf(mv.w)};
+  }
+  if (threadIdx.x < (n & 3)) {
+    int j = n4 * 4 + threadIdx.x;
+    float v = m[j] - coef * g[j];
+    m[j] = v;
+    if (sh) sh[j] = f2bf(v);
+  }
+}
+
+void launch_sgd_mt(const int64_t* desc, int nchunk, const float* norm2,
+                   float max_norm, float lr, float grad_scale,
+                   hipStream_t stream) {
+  hipLaunchKernelGGL(sgd_mt_kernel, dim3(nchunk), dim3(256), 0, stream, desc,
+                     norm2, max_norm, lr, grad_scale);
+}
+
 // master -= lr * grad_scale * clip(norm) * grad; shadow/bf16 rewritten in
 // the same pass. `shadow` may be null (biases stay fp32).
 __global__ void sgd_update_kernel(float* __restrict__ master,

@@ -377,6 +377,18 @@ static void lsm_nll_bwd(const torch::Tensor& scores, const torch::Tensor& lse,
                      current_stream());
 }
 
+static void norm2_mt(const torch::Tensor& desc, torch::Tensor& accum) {
+  launch_norm2_mt(desc.data_ptr<int64_t>(), desc.size(0), f_ptr_mut(accum),
+                  current_stream());
+}
+
+static void sgd_mt(const torch::Tensor& desc, const torch::Tensor& norm2,
+                   double max_norm, double lr, double grad_scale) {
+  launch_sgd_mt(desc.data_ptr<int64_t>(), desc.size(0), f_ptr(norm2),
+                (float)max_norm, (float)lr, (float)grad_scale,
+                current_stream());
+}
+
 static void norm2_accum(const torch::Tensor& g, torch::Tensor& accum) {
   launch_norm2_accum(f_ptr(g), g.numel(), f_ptr_mut(accum), current_stream());
 }
@@ -420,6 +432,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lsm_nll_fwd", &zamd::lsm_nll_fwd);
   m.def("lsm_nll_bwd", &zamd::lsm_nll_bwd);
   m.def("norm2_accum", &zamd::norm2_accum);
+  m.def("norm2_mt", &zamd::norm2_mt);
+  m.def("sgd_mt", &zamd::sgd_mt);
   m.def("sgd_update", &zamd::sgd_update);
   m.def("transpose_bf16", &zamd::transpose_bf16);
   m.def("colsum_bf16", &zamd::colsum_bf16);

@@ -64,6 +64,11 @@ void launch_lsm_nll_bwd(const float* scores, const float* lse,
                         float* dscores, int N, int V, hipStream_t stream);
 void launch_norm2_accum(const float* g, int64_t n, float* accum,
                         hipStream_t stream);
+void launch_norm2_mt(const int64_t* desc, int nchunk, float* accum,
+                     hipStream_t stream);
+void launch_sgd_mt(const int64_t* desc, int nchunk, const float* norm2,
+                   float max_norm, float lr, float grad_scale,
+                   hipStream_t stream);
 void launch_sgd_update(float* master, const float* grad, bf16* shadow,
                        const float* norm2, float max_norm, float lr,
                        float grad_scale, int64_t n, hipStream_t stream);

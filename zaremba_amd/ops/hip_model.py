@@ -195,18 +195,41 @@ class HipModel:
         e = self.e
         self.norm2.zero_()
         params = [p for p in m.parameters() if p.grad is not None]
-        for p in params:
-            g = p.grad
-            e.norm2_accum(g.reshape(-1), self.norm2)
         shadow_of = {id(m.embed.W): self.emb_W, id(m.fc.W): self.fc.W}
         for rt, layer in zip(self.layers, m.rnns):
             shadow_of[id(layer.W_x)] = rt.Wx
             shadow_of[id(layer.W_h)] = rt.Wh
-        for p in params:
-            sh = shadow_of.get(id(p))
-            e.sgd_update(p.data.view(-1), p.grad.reshape(-1),
-                         sh.view(-1) if sh is not None else None,
-                         self.norm2, max_norm, lr, grad_scale)
+        # Chunked multi-tensor descriptors: one norm launch + one update
+        # launch for all params. Grad tensors are fresh per step, but the
+        # caching allocator cycles through a small set of addresses, so
+        # descriptor tensors are memoized per pointer signature.
+        key = tuple(p.grad.data_ptr() for p in params)
+        cache = getattr(self, "_mt_cache", None)
+        if cache is None:
+            cache = self._mt_cache = {}
+        descs = cache.get(key)
+        if descs is None:
+            CH = 65536
+            nd, sd = [], []
+            for p in params:
+                mp, gp = p.data.data_ptr(), p.grad.data_ptr()
+                sh = shadow_of.get(id(p))
+                sp = sh.data_ptr() if sh is not None else 0
+                n = p.numel()
+                for off in range(0, n, CH):
+                    ln = min(CH, n - off)
+                    nd.append((gp + off * 4, ln))
+                    sd.append((mp + off * 4, gp + off * 4,
+                               sp + off * 2 if sp else 0, ln))
+            descs = (
+                torch.tensor(nd, dtype=torch.int64).to(self.device),
+                torch.tensor(sd, dtype=torch.int64).to(self.device),
+            )
+            if len(cache) > 32:
+                cache.clear()
+            cache[key] = descs
+        e.norm2_mt(descs[0], self.norm2)
+        e.sgd_mt(descs[1], self.norm2, max_norm, lr, grad_scale)
         # refresh the derived shadows (transposes + packs + folded biases)
         for rt, layer in zip(self.layers, m.rnns):
             e.transpose_bf16(rt.Wx, rt.WxT)
