@@ -93,16 +93,25 @@ int main() {
     (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
     printf("census nb=%3d: full=%u timeout=%u\n", nb, res[0], res[1]);
   }
-  // barrier probe at the largest resident size
-  for (int nb : {250, 248, 240}) {
-    (void)hipMemset(buf, 0, 4096);
-    hipLaunchKernelGGL(barrier_probe_kernel, dim3(nb), dim3(256), lds, 0,
-                       buf, buf + 64, buf + 65, 200, nb);
-    (void)hipDeviceSynchronize();
-    unsigned int res[2];
-    (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
-    printf("barrier nb=%3d iters=200: fail=%u done=%u\n", nb, res[0],
-           res[1]);
+  // barrier probe at the largest resident size (timed)
+  for (int nb : {250, 125}) {
+    for (int iters : {1000}) {
+      (void)hipMemset(buf, 0, 4096);
+      hipEvent_t e0, e1;
+      (void)hipEventCreate(&e0);
+      (void)hipEventCreate(&e1);
+      (void)hipEventRecord(e0, 0);
+      hipLaunchKernelGGL(barrier_probe_kernel, dim3(nb), dim3(256), lds, 0,
+                         buf, buf + 64, buf + 65, iters, nb);
+      (void)hipEventRecord(e1, 0);
+      (void)hipDeviceSynchronize();
+      float ms = 0;
+      (void)hipEventElapsedTime(&ms, e0, e1);
+      unsigned int res[2];
+      (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
+      printf("barrier nb=%3d iters=%d: fail=%u done=%u  %.2f us/barrier\n",
+             nb, iters, res[0], res[1], ms * 1000.f / iters);
+    }
   }
   return 0;
 }
