@@ -37,7 +37,14 @@ def data_init(data_dir: str = "./data"):
     Returns ``(trn, vld, tst, vocab_size)`` where each split is an
     ``np.ndarray`` of shape ``[N, 1]`` int64 (reference ``main.py:44-59``).
     """
-    trn = read_tokens(os.path.join(data_dir, "ptb.train.txt"))
+    train_path = os.path.join(data_dir, "ptb.train.txt")
+    if not os.path.exists(train_path):
+        raise FileNotFoundError(
+            f"{train_path} not found. The PTB train split is not "
+            "redistributable here (it is a missing blob in the reference "
+            "repo too); place your own copy in the data dir, or run with "
+            "--data synthetic for throughput/benchmark work.")
+    trn = read_tokens(train_path)
     vld = read_tokens(os.path.join(data_dir, "ptb.valid.txt"))
     tst = read_tokens(os.path.join(data_dir, "ptb.test.txt"))
     words = sorted(set(trn))
