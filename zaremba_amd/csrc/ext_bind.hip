@@ -106,7 +106,8 @@ static bool g_use_persistent_bwd = false;  // measured slower; kept for A/B
 // element per thread: B*HS <= 256, H even, B <= 32.
 static bool persistent_ok(int B, int H) {
   return g_use_persistent && B <= 32 && (H % 2) == 0 &&
-         B * persistent_hs(H) <= 256;
+         B * persistent_hs(H) <= 256 &&
+         persistent_fwd_lds(B, H) <= 160 * 1024;
 }
 
 static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,

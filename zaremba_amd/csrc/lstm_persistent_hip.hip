@@ -44,8 +44,20 @@ constexpr int PCELL_THREADS = 256;
 constexpr int REC_CH = 6;  // i,f,o,n,tanh_c_new,c_prev channels
 
 // HS must be even; NB = ceil(H/HS) <= 250 so the grid is co-resident.
+// Prefer HS=8 (NB=188 at H=1500): the grid barrier is the forward's
+// dominant per-step cost and scales with workgroup count (census probe:
+// 9.9 us @250 WGs, 5.1 @125).
 int persistent_hs(int H) {
-  return 2 * cdiv(H, 2 * 250);
+  int hs = 2 * cdiv(H, 2 * 250);
+  return hs < 8 ? 8 : hs;
+}
+
+// dynamic-LDS bytes the forward kernel needs (B <= 32)
+size_t persistent_fwd_lds(int B, int H) {
+  const int HS = persistent_hs(H);
+  const int KPAD = ((H + 31) / 32) * 32 + 8;
+  return (size_t)(4 * HS + B) * KPAD * 2 + (size_t)4 * B * 16 * 4 +
+         (size_t)B * HS * 2 + 16;
 }
 
 #define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
@@ -112,8 +124,8 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* Ws = reinterpret_cast<bf16*>(smem);            // [4*HS][KPAD]
   bf16* hs = Ws + (int64_t)4 * HS * KPAD;              // [B][KPAD]
-  float* gbuf = reinterpret_cast<float*>(hs + (int64_t)B * KPAD);
-  bf16* hbuf = reinterpret_cast<bf16*>(gbuf + 4 * MAXB * 16);  // [B][HS]
+  float* gbuf = reinterpret_cast<float*>(hs + (int64_t)B * KPAD);  // [4][B][16]
+  bf16* hbuf = reinterpret_cast<bf16*>(gbuf + 4 * B * 16);  // [B][HS]
 
   const int NB = (H + HS - 1) / HS;
   const int grp = blockIdx.x & 7;
@@ -223,8 +235,8 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      gbuf[(g * MAXB + fr0 + r) * 16 + lm] = acc0[r];
-      gbuf[(g * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+      if (fr0 + r < B) gbuf[(g * B + fr0 + r) * 16 + lm] = acc0[r];
+      if (16 + fr0 + r < B) gbuf[(g * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
     }
     __syncthreads();
 
@@ -233,10 +245,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       const int b = own_b, jj = own_jj;
       const int j = j0 + jj;
       const int64_t gxbase = ((int64_t)t * B + b) * 4 * H + j;
-      float gi = gbuf[(0 * MAXB + b) * 16 + jj] + bf2f(gx[gxbase + 0 * H]);
-      float gf = gbuf[(1 * MAXB + b) * 16 + jj] + bf2f(gx[gxbase + 1 * H]);
-      float go = gbuf[(2 * MAXB + b) * 16 + jj] + bf2f(gx[gxbase + 2 * H]);
-      float gn = gbuf[(3 * MAXB + b) * 16 + jj] + bf2f(gx[gxbase + 3 * H]);
+      float gi = gbuf[(0 * B + b) * 16 + jj] + bf2f(gx[gxbase + 0 * H]);
+      float gf = gbuf[(1 * B + b) * 16 + jj] + bf2f(gx[gxbase + 1 * H]);
+      float go = gbuf[(2 * B + b) * 16 + jj] + bf2f(gx[gxbase + 2 * H]);
+      float gn = gbuf[(3 * B + b) * 16 + jj] + bf2f(gx[gxbase + 3 * H]);
       float i_ = 1.f / (1.f + __expf(-gi));
       float f_ = 1.f / (1.f + __expf(-gf));
       float o_ = 1.f / (1.f + __expf(-go));
@@ -282,8 +294,7 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
   const int NB = cdiv(H, HS);
   const int KS = (H + 31) / 32;
   const int KPAD = KS * 32 + 8;
-  size_t lds = (size_t)(4 * HS + B) * KPAD * 2 + 4 * 32 * 16 * sizeof(float) +
-               (size_t)B * HS * 2 + 16;
+  size_t lds = persistent_fwd_lds(B, H);
   hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32>), dim3(NB),
                      dim3(PCELL_THREADS), lds, stream, gx, W_h, h_all, c_all,
                      rec, pstate, abort_flag, T, B, H, HS);

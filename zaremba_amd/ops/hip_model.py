@@ -20,7 +20,7 @@ from typing import List, Optional
 
 import torch
 
-from .. import _C
+from .. import _C  # noqa
 from .hip_ops import DropoutFn, EmbeddingFn, LinearFn, LstmLayerFn
 
 
@@ -48,7 +48,7 @@ class _LayerWorkspace:
                                    device=device)
         # persistent-kernel state: block records [T][NB][B][6][HS],
         # barrier words (re-zeroed per call by the driver) + abort flag
-        hs = 2 * ((H + 499) // 500)
+        hs = _C.ext().persistent_hs(H)
         nb = (H + hs - 1) // hs
         self.rec = torch.zeros(T * nb * B * 6 * hs, dtype=bf, device=device)
         self.hgran = torch.zeros(32, dtype=torch.int64, device=device)
