@@ -207,12 +207,14 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
   const int64_t gstep = (int64_t)B * 4 * H;
   const int HSp = persistent_hs(H);
   const int64_t rstep = (int64_t)((H + HSp - 1) / HSp) * B * 6 * HSp;
+  float* dh2 = dh_rec + (int64_t)B * H;  // second K-slice partial
   for (int t = T - 1; t >= 0; --t) {
     launch_lstm_cell_bwd_elt(dY + t * hstep,
-                             (t == T - 1) ? nullptr : dh_rec, dc,
+                             (t == T - 1) ? nullptr : dh_rec, dh2, dc,
                              rec + t * rstep, dG + t * gstep, dG_pack,
                              B, H, HSp, stream);
-    launch_smallm_packed_nt(dG_pack, WT_pack, dh_rec, B, H, 4 * H, stream);
+    launch_smallm_packed_nt(dG_pack, WT_pack, dh_rec, dh2, B, H, 4 * H,
+                            stream);
   }
 }
 
@@ -325,8 +327,8 @@ static void smallm_gemm_nt(const torch::Tensor& A, const torch::Tensor& B,
   TORCH_CHECK(M <= 32, "smallm gemm requires M <= 32");
   auto ap = pack_a_tmp(A);
   auto wp = pack_w_tmp(B, N, 1, K);
-  launch_smallm_packed_nt(bf_ptr(ap), bf_ptr(wp), f_ptr_mut(C), M, N, K,
-                          current_stream());
+  launch_smallm_packed_nt(bf_ptr(ap), bf_ptr(wp), f_ptr_mut(C), nullptr, M,
+                          N, K, current_stream());
 }
 
 // ---------------------------------------------------------------------------

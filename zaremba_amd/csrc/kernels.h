@@ -26,11 +26,12 @@ void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
                           bf16* h_pack_out, float* c_out, bf16* gates_out,
                           bf16* rec, int B, int H, int HSp,
                           hipStream_t stream);
-void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
+void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec,
+                              const float* dh_rec2, float* dc,
                               const bf16* rec, bf16* dG, bf16* dG_pack,
                               int B, int H, int HSp, hipStream_t stream);
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
-                             float* C, int M, int N, int K,
+                             float* C, float* C2, int M, int N, int K,
                              hipStream_t stream);
 
 // lstm_persistent.hip — one launch for a whole layer unroll

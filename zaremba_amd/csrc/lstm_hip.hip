@@ -223,9 +223,9 @@ void launch_lstm_cell_fwd(const bf16* h_pack, const float* c_prev,
 // ---------------------------------------------------------------------------
 __global__ void lstm_cell_bwd_elt_kernel(
     const bf16* __restrict__ dy, const float* __restrict__ dh_rec,
-    float* __restrict__ dc, const bf16* __restrict__ rec,
-    bf16* __restrict__ dG, bf16* __restrict__ dG_pack, int B, int H,
-    int HSp) {
+    const float* __restrict__ dh_rec2, float* __restrict__ dc,
+    const bf16* __restrict__ rec, bf16* __restrict__ dG,
+    bf16* __restrict__ dG_pack, int B, int H, int HSp) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= B * H) return;
   int b = idx / H, j = idx % H;
@@ -238,7 +238,7 @@ __global__ void lstm_cell_bwd_elt_kernel(
   float n_ = bf2f(r[3 * HSp + jr]);
   float tc = bf2f(r[4 * HSp + jr]);
   float cprev = bf2f(r[5 * HSp + jr]);
-  float dh = bf2f(dy[idx]) + (dh_rec ? dh_rec[idx] : 0.f);
+  float dh = bf2f(dy[idx]) + (dh_rec ? dh_rec[idx] + dh_rec2[idx] : 0.f);
   float do_ = dh * tc;
   float dct = dc[idx] + dh * o_ * (1.f - tc * tc);
   float di = dct * n_;
@@ -257,12 +257,14 @@ __global__ void lstm_cell_bwd_elt_kernel(
   dc[idx] = dct * f_;
 }
 
-void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
+void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec,
+                              const float* dh_rec2, float* dc,
                               const bf16* rec, bf16* dG, bf16* dG_pack,
                               int B, int H, int HSp, hipStream_t stream) {
   int n = B * H;
   hipLaunchKernelGGL(lstm_cell_bwd_elt_kernel, dim3(cdiv(n, 256)), dim3(256),
-                     0, stream, dy, dh_rec, dc, rec, dG, dG_pack, B, H, HSp);
+                     0, stream, dy, dh_rec, dh_rec2, dc, rec, dG, dG_pack, B,
+                     H, HSp);
 }
 
 // ---------------------------------------------------------------------------
@@ -273,30 +275,35 @@ void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec, float* dc,
 // ([ceil(N/16)][KS][64][8]). Grid = ceil(N/16); the 4 waves round-robin
 // the K steps; one LDS reduction at the end. Fully uniform — pads are
 // zero in both packs.
-template <int MAXB>
+template <int MAXB, int SK>
 __global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
     const bf16* __restrict__ A_pack, const bf16* __restrict__ W_pack,
-    float* __restrict__ C, int M, int N, int K) {
+    float* __restrict__ C, float* __restrict__ C2, int M, int N, int K) {
   __shared__ float red[4 * MAXB * 16];
 
-  const int n0 = blockIdx.x * 16;
+  const int nbn = (N + 15) / 16;
+  const int n0 = (blockIdx.x % nbn) * 16;
+  const int sk = blockIdx.x / nbn;           // K slice (0..SK-1)
   const int w = wave_id();
   const int l = lane_id();
   const int lm = l & 15;
   const int KS = (K + 31) / 32;
+  const int KH = (KS + SK - 1) / SK;         // steps per slice
+  const int ks0 = sk * KH;
+  const int ks1 = min(ks0 + KH, KS);
 
   const bf16x8* pa = reinterpret_cast<const bf16x8*>(A_pack) + l;
   const bf16x8* pw = reinterpret_cast<const bf16x8*>(W_pack) +
-                     (int64_t)blockIdx.x * KS * 64 + l;
+                     (int64_t)(blockIdx.x % nbn) * KS * 64 + l;
 
   f32x4 acc0 = {}, acc1 = {};
-  const int nown = (KS - w + 3) / 4;  // steps owned by this wave (w + 4i)
+  const int nown = (ks1 - ks0 - w + 3) / 4;  // owned steps: ks0 + w + 4i
   int i = 0;
   for (; i + 8 <= nown; i += 8) {
     bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const int ks = w + 4 * (i + u);
+      const int ks = ks0 + w + 4 * (i + u);
       a0v[u] = pa[ks * 128];
       a1v[u] = pa[ks * 128 + 64];
       bwv[u] = pw[ks * 64];
@@ -308,7 +315,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
     }
   }
   for (; i < nown; ++i) {
-    const int ks = w + 4 * i;
+    const int ks = ks0 + w + 4 * i;
     bf16x8 a0v = pa[ks * 128];
     bf16x8 a1v = pa[ks * 128 + 64];
     bf16x8 bwv = pw[ks * 64];
@@ -323,21 +330,28 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
     red[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
   }
   __syncthreads();
+  float* out = (SK > 1 && sk == 1) ? C2 : C;
   for (int idx = threadIdx.x; idx < M * 16; idx += CELL_THREADS) {
     const int b = idx / 16, jj = idx % 16;
     if (n0 + jj >= N) continue;
     float v = red[(0 * MAXB + b) * 16 + jj] + red[(1 * MAXB + b) * 16 + jj] +
               red[(2 * MAXB + b) * 16 + jj] + red[(3 * MAXB + b) * 16 + jj];
-    C[(int64_t)b * N + n0 + jj] = v;
+    out[(int64_t)b * N + n0 + jj] = v;
   }
 }
 
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
-                             float* C, int M, int N, int K,
+                             float* C, float* C2, int M, int N, int K,
                              hipStream_t stream) {
-  hipLaunchKernelGGL((smallm_packed_nt_kernel<32>), dim3(cdiv(N, 16)),
-                     dim3(CELL_THREADS), 0, stream, A_pack, W_pack, C, M, N,
-                     K);
+  if (C2) {
+    hipLaunchKernelGGL((smallm_packed_nt_kernel<32, 2>),
+                       dim3(cdiv(N, 16) * 2), dim3(CELL_THREADS), 0, stream,
+                       A_pack, W_pack, C, C2, M, N, K);
+  } else {
+    hipLaunchKernelGGL((smallm_packed_nt_kernel<32, 1>), dim3(cdiv(N, 16)),
+                       dim3(CELL_THREADS), 0, stream, A_pack, W_pack, C,
+                       nullptr, M, N, K);
+  }
 }
 
 }  // namespace zamd

@@ -38,7 +38,7 @@ class _LayerWorkspace:
         self.gates = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
         self.dY = torch.zeros(T, B, H, dtype=bf, device=device)
         self.dG = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
-        self.dh_rec = torch.zeros(B, H, dtype=f32, device=device)
+        self.dh_rec = torch.zeros(2, B, H, dtype=f32, device=device)
         self.dc = torch.zeros(B, H, dtype=f32, device=device)
         # fragment-packed workspaces (zero-prefilled: pad rows/K-tails
         # must read as 0.0 in the packed cell kernels)
