@@ -26,6 +26,7 @@ import numpy as np
 import torch
 
 from .ops import functional as F_ref
+from .utils.profiling import trace_range
 
 
 def nll_loss(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
@@ -145,18 +146,23 @@ def train(
             if dp is not None:
                 dp.zero_grad()
             states = model.detach(states)
-            scores, states = model(x, states)
-            loss = nll_loss(scores, y.to(scores.device))
-            loss.backward()
+            with trace_range("forward"):
+                scores, states = model(x, states)
+            with trace_range("loss"):
+                loss = nll_loss(scores, y.to(scores.device))
+            with trace_range("backward"):
+                loss.backward()
             if dp is not None:
-                dp.finalize_backward()
+                with trace_range("allreduce_join"):
+                    dp.finalize_backward()
             # Under DP each rank holds the SUM of per-rank grads after the
             # all-reduce. The reference loss SUMS over the batch dimension
             # (main.py:82-84), so summed grads are exactly the gradients
             # the reference would compute at batch_size = world * B — the
             # "global batch 160" semantics of BASELINE config 4. No
             # averaging (grad_scale stays 1).
-            norm = sgd_step(model, lr, max_norm)
+            with trace_range("clip_sgd"):
+                norm = sgd_step(model, lr, max_norm)
             if i % cadence == 0 and is_rank0:
                 toc = timeit.default_timer()
                 norm_v = float(norm)
