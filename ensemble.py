@@ -81,16 +81,8 @@ def build_parser():
 
 
 def load_data(args):
-    if args.data.startswith("synthetic"):
-        vocab = 10000
-        if ":" in args.data:
-            for kv in args.data.split(":", 1)[1].split(","):
-                k, v = kv.split("=")
-                if k == "vocab":
-                    vocab = int(v)
-        return zdata.synthetic_init(vocab_size=vocab,
-                                    seed=args.seed if args.seed is not None else 1234)
-    return zdata.data_init(args.data_dir)
+    from main import load_data as _ld
+    return _ld(args)
 
 
 def train_one(args, vocab_size, data, model_num):

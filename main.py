@@ -66,8 +66,10 @@ def build_parser():
                         default="bf16",
                         help="GPU compute dtype (fp32 master weights either way).")
     parser.add_argument("--data", type=str, default="ptb",
-                        help="'ptb' or 'synthetic[:vocab=N]' (random tokens, "
-                             "PTB-shaped; used when ptb.train.txt is absent).")
+                        help="'ptb', 'synthetic[:vocab=N]' (uniform random, "
+                             "PTB-shaped) or 'synthetic_markov[:vocab=N,"
+                             "branch=K]' (learnable order-1 chain, optimal "
+                             "perplexity = K).")
     parser.add_argument("--data_dir", type=str, default="./data",
                         help="Directory holding ptb.{train,valid,test}.txt.")
     parser.add_argument("--save", type=str, default=None,
@@ -96,14 +98,16 @@ def setdevice(args, plural=False):
 
 def load_data(args):
     if args.data.startswith("synthetic"):
-        vocab = 10000
+        opts = {"vocab": 10000, "branch": 20}
         if ":" in args.data:
             for kv in args.data.split(":", 1)[1].split(","):
                 k, v = kv.split("=")
-                if k == "vocab":
-                    vocab = int(v)
-        return zdata.synthetic_init(vocab_size=vocab,
-                                    seed=args.seed if args.seed is not None else 1234)
+                opts[k] = int(v)
+        seed = args.seed if args.seed is not None else 1234
+        if args.data.startswith("synthetic_markov"):
+            return zdata.synthetic_markov_init(
+                vocab_size=opts["vocab"], branch=opts["branch"], seed=seed)
+        return zdata.synthetic_init(vocab_size=opts["vocab"], seed=seed)
     return zdata.data_init(args.data_dir)
 
 

@@ -69,3 +69,18 @@ def test_shard_stream_disjoint():
     cat = np.concatenate([s.reshape(-1) for s in shards])
     assert len(cat) == 100
     assert len(set(cat.tolist())) == 100
+
+
+def test_synthetic_markov_learnable_structure():
+    trn, vld, tst, v = zdata.synthetic_markov_init(
+        vocab_size=200, branch=5, train_tokens=3000, valid_tokens=300,
+        test_tokens=300)
+    assert v == 200
+    # every observed bigram must be one of the 5 successors of its
+    # predecessor (the property that makes perplexity=branch achievable)
+    flat = trn.reshape(-1)
+    import collections
+    succ = collections.defaultdict(set)
+    for a, b in zip(flat[:-1], flat[1:]):
+        succ[int(a)].add(int(b))
+    assert max(len(s) for s in succ.values()) <= 5

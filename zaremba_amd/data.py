@@ -77,6 +77,39 @@ def synthetic_init(
     return trn, vld, tst, vocab_size
 
 
+def synthetic_markov_init(
+    vocab_size: int = 10000,
+    branch: int = 20,
+    train_tokens: int = 929589,
+    valid_tokens: int = 73760,
+    test_tokens: int = 82430,
+    seed: int = 1234,
+):
+    """A LEARNABLE synthetic corpus: an order-1 Markov chain where every
+    token has exactly ``branch`` equiprobable successors. The optimal
+    perplexity is ``branch`` — a trained model approaching it end-to-end
+    validates the whole bf16 kernel/training stack (the real PTB train
+    split is a missing blob upstream)."""
+    rng = np.random.default_rng(seed)
+    successors = rng.integers(0, vocab_size, size=(vocab_size, branch),
+                              dtype=np.int64)
+
+    def walk(n, state):
+        out = np.empty(n, dtype=np.int64)
+        choices = rng.integers(0, branch, size=n)
+        for i in range(n):
+            state = successors[state, choices[i]]
+            out[i] = state
+        return out, state
+
+    state = 0
+    trn, state = walk(train_tokens, state)
+    vld, state = walk(valid_tokens, state)
+    tst, state = walk(test_tokens, state)
+    return (trn.reshape(-1, 1), vld.reshape(-1, 1), tst.reshape(-1, 1),
+            vocab_size)
+
+
 def minibatch(
     data: np.ndarray, batch_size: int, seq_length: int
 ) -> List[Tuple[torch.Tensor, torch.Tensor]]:
