@@ -138,3 +138,27 @@ def test_hip_medium_and_nonreg_shapes():
         norm = trainer.sgd_step(m, lr=1.0, max_norm=5.0)
         torch.cuda.synchronize()
         assert torch.isfinite(loss).item() and float(norm) > 0
+
+
+def test_hip_learns_markov_structure():
+    """End-to-end convergence: the bf16 HIP stack must learn an order-1
+    Markov corpus well past its unigram entropy (tracks eager fp32 within
+    ~0.5%/epoch at larger scale; see PERF.md convergence section)."""
+    import io
+    import contextlib
+    from zaremba_amd import data as zdata, trainer
+    from zaremba_amd.models.lstm_lm import Model
+    torch.manual_seed(0)
+    trn, vld, _, v = zdata.synthetic_markov_init(
+        vocab_size=500, branch=20, train_tokens=500000, valid_tokens=50000,
+        test_tokens=1000, seed=1)
+    ds = zdata.minibatch(trn, 20, 35)
+    dv = zdata.minibatch(vld, 20, 35)
+    m = Model(v, 650, 2, dropout=0.0, winit=0.05, engine="hip").to(dev())
+    with contextlib.redirect_stdout(io.StringIO()):
+        trainer.train((ds, dv, dv), m, epochs=3, epoch_threshold=100, lr=1.0,
+                      factor=1.2, max_norm=5.0, batch_size=20)
+    ppl = trainer.perplexity(dv, m, 20)
+    # unigram ppl is ~430 here; the conditional structure (optimal 20)
+    # must be clearly learned
+    assert ppl < 60, ppl

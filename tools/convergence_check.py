@@ -9,10 +9,15 @@ from zaremba_amd.models.lstm_lm import Model
 
 epochs = int(sys.argv[1]) if len(sys.argv) > 1 else 12
 engine = sys.argv[2] if len(sys.argv) > 2 else "auto"
+vocab = int(sys.argv[3]) if len(sys.argv) > 3 else 2000
+# ~50 observations per (state, successor) transition: learnable in a few
+# epochs (vocab 10000 with only 400K tokens gave ~2 obs/transition and
+# plateaued at the unigram entropy for EVERY engine, including eager fp32)
+tokens = vocab * 20 * 50
 torch.manual_seed(0)
 trn, vld, tst, v = zdata.synthetic_markov_init(
-    vocab_size=10000, branch=20, train_tokens=400000, valid_tokens=40000,
-    test_tokens=40000, seed=1)
+    vocab_size=vocab, branch=20, train_tokens=tokens,
+    valid_tokens=tokens // 10, test_tokens=tokens // 10, seed=1)
 ds = zdata.minibatch(trn, 20, 35)
 dv = zdata.minibatch(vld, 20, 35)
 dev = "cuda" if torch.cuda.is_available() else "cpu"
