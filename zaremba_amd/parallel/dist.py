@@ -21,10 +21,17 @@ def env_world_size() -> int:
 
 
 def maybe_init():
-    """Initialize the default process group when launched by torchrun."""
+    """Initialize the default process group when launched by torchrun.
+
+    Test overrides: ZAREMBA_AMD_PG_BACKEND forces the backend (e.g. gloo
+    to exercise the DP code path on one GPU) and ZAREMBA_AMD_ONE_GPU=1
+    pins every rank to device 0 (multi-rank smoke tests on a single-GPU
+    box)."""
     if env_world_size() <= 1 or td.is_initialized():
         return
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get(
+        "ZAREMBA_AMD_PG_BACKEND",
+        "nccl" if torch.cuda.is_available() else "gloo")
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank())
     td.init_process_group(backend=backend,
@@ -44,6 +51,8 @@ def world_size() -> int:
 
 
 def local_rank() -> int:
+    if os.environ.get("ZAREMBA_AMD_ONE_GPU", "0") == "1":
+        return 0
     return int(os.environ.get("LOCAL_RANK", "0"))
 
 
