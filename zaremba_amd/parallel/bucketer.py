@@ -27,11 +27,17 @@ import torch.distributed as td
 
 
 class _Bucket:
-    def __init__(self, params: List[torch.nn.Parameter]):
+    def __init__(self, params: List[torch.nn.Parameter], comm_dtype):
         self.params = params
         total = sum(p.numel() for p in params)
         dev = params[0].device
         self.flat = torch.zeros(total, dtype=torch.float32, device=dev)
+        # wire format: bf16 halves the xGMI bytes (266 MB fp32 grads for
+        # the Large model would otherwise rival the 3.4 ms compute step
+        # at DP8); fp32 wire via ZAREMBA_AMD_ALLREDUCE_FP32=1
+        self.comm_dtype = comm_dtype
+        self.comm = (self.flat if comm_dtype == torch.float32 else
+                     torch.zeros(total, dtype=comm_dtype, device=dev))
         offset = 0
         for p in params:
             n = p.numel()
@@ -46,9 +52,16 @@ class _Bucket:
 
 
 class GradBucketer:
-    def __init__(self, model: torch.nn.Module, bucket_bytes: int = 25 << 20):
+    def __init__(self, model: torch.nn.Module, bucket_bytes: int = 25 << 20,
+                 comm_dtype=None):
         if not (td.is_available() and td.is_initialized()):
             raise RuntimeError("GradBucketer requires an initialized process group")
+        if comm_dtype is None:
+            import os
+            comm_dtype = (torch.float32
+                          if os.environ.get("ZAREMBA_AMD_ALLREDUCE_FP32") == "1"
+                          or not torch.cuda.is_available()
+                          else torch.bfloat16)
         self.world_size = td.get_world_size()
         params = [p for p in model.parameters() if p.requires_grad]
         params.reverse()  # backward completion order
@@ -59,10 +72,10 @@ class GradBucketer:
             cur.append(p)
             cur_bytes += p.numel() * 4
             if cur_bytes >= bucket_bytes:
-                self.buckets.append(_Bucket(cur))
+                self.buckets.append(_Bucket(cur, comm_dtype))
                 cur, cur_bytes = [], 0
         if cur:
-            self.buckets.append(_Bucket(cur))
+            self.buckets.append(_Bucket(cur, comm_dtype))
         self._by_param = {}
         self._hooks = []
         for b in self.buckets:
@@ -75,7 +88,9 @@ class GradBucketer:
         b = self._by_param[param]
         b.pending -= 1
         if b.pending == 0:
-            b.work = td.all_reduce(b.flat, op=td.ReduceOp.SUM, async_op=True)
+            if b.comm is not b.flat:
+                b.comm.copy_(b.flat)
+            b.work = td.all_reduce(b.comm, op=td.ReduceOp.SUM, async_op=True)
 
     def zero_grad(self):
         for b in self.buckets:
@@ -85,10 +100,16 @@ class GradBucketer:
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()
+                if b.comm is not b.flat:
+                    b.flat.copy_(b.comm)
             elif b.pending != 0:
                 # A parameter produced no grad this step (should not happen in
                 # this model); reduce anyway so replicas stay in sync.
-                td.all_reduce(b.flat, op=td.ReduceOp.SUM)
+                if b.comm is not b.flat:
+                    b.comm.copy_(b.flat)
+                td.all_reduce(b.comm, op=td.ReduceOp.SUM)
+                if b.comm is not b.flat:
+                    b.flat.copy_(b.comm)
 
     def detach_hooks(self):
         for h in self._hooks:
