@@ -175,6 +175,19 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
   const int lk = (l >> 4) * 8;
 
   for (int t = 0; t < T; ++t) {
+    // Prefetch this thread's gx slice BEFORE the barrier: the loads are
+    // independent of h_t, and gx[t] is LLC-cold scattered data whose
+    // latency otherwise lands inside the pointwise phase.
+    bf16 gxi = (bf16)0.f, gxf = (bf16)0.f, gxo = (bf16)0.f,
+         gxn = (bf16)0.f;
+    if (own) {
+      const int64_t gxb =
+          ((int64_t)t * B + own_b) * 4 * H + j0 + own_jj;
+      gxi = gx[gxb + 0 * H];
+      gxf = gx[gxb + 1 * H];
+      gxo = gx[gxb + 2 * H];
+      gxn = gx[gxb + 3 * H];
+    }
     if (t > 0) {
       if (!xcd_grid_barrier(pstate, grp, nbg, ngroups, (unsigned int)t,
                             abort_flag))
@@ -245,10 +258,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       const int b = own_b, jj = own_jj;
       const int j = j0 + jj;
       const int64_t gxbase = ((int64_t)t * B + b) * 4 * H + j;
-      float gi = gbuf[(0 * B + b) * 16 + jj] + bf2f(gx[gxbase + 0 * H]);
-      float gf = gbuf[(1 * B + b) * 16 + jj] + bf2f(gx[gxbase + 1 * H]);
-      float go = gbuf[(2 * B + b) * 16 + jj] + bf2f(gx[gxbase + 2 * H]);
-      float gn = gbuf[(3 * B + b) * 16 + jj] + bf2f(gx[gxbase + 3 * H]);
+      float gi = gbuf[(0 * B + b) * 16 + jj] + bf2f(gxi);
+      float gf = gbuf[(1 * B + b) * 16 + jj] + bf2f(gxf);
+      float go = gbuf[(2 * B + b) * 16 + jj] + bf2f(gxo);
+      float gn = gbuf[(3 * B + b) * 16 + jj] + bf2f(gxn);
       float i_ = 1.f / (1.f + __expf(-gi));
       float f_ = 1.f / (1.f + __expf(-gf));
       float o_ = 1.f / (1.f + __expf(-go));
