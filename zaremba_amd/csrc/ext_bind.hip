@@ -100,7 +100,7 @@ struct GraphCache {
 static GraphCache g_fwd_graphs, g_bwd_graphs;
 static bool g_use_graphs = true;
 static bool g_use_persistent = true;
-static bool g_use_persistent_bwd = false;  // measured slower; kept for A/B
+static bool g_use_persistent_bwd = false;  // see PERF.md: broadcast amplification loses
 
 // The persistent forward needs every block co-resident and one cell
 // element per thread: B*HS <= 256, H even, B <= 32.
@@ -196,9 +196,10 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
                               float* dc, unsigned long long* hgran,
                               unsigned int* abort_flag, int T, int B, int H,
                               hipStream_t stream) {
-  if (g_use_persistent_bwd && persistent_ok(B, H)) {
+  if (g_use_persistent_bwd && persistent_ok(B, H) &&
+      persistent_bwd_lds(B, H) <= 160 * 1024) {
     HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
-    launch_lstm_persistent_bwd(dY, rec, W_h_T, dG,
+    launch_lstm_persistent_bwd(dY, rec, W_h_T, dG, dG_pack,
                                reinterpret_cast<unsigned int*>(hgran),
                                abort_flag, T, B, H, stream);
     return;
@@ -242,7 +243,8 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
   auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
   auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
   auto stream = current_stream();
-  if (!g_use_graphs || (g_use_persistent_bwd && persistent_ok(B, H))) {
+  if (!g_use_graphs || (g_use_persistent_bwd && persistent_ok(B, H) &&
+                        persistent_bwd_lds(B, H) <= 160 * 1024)) {
     lstm_seq_bwd_body(dyp, gp, rp, cp, whtp, wtp, dgp, dgpk, dhp, dcp, hg,
                       ab, T, B, H, stream);
     return;

@@ -42,8 +42,9 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
                                 unsigned int* pstate,
                                 unsigned int* abort_flag, int T, int B,
                                 int H, hipStream_t stream);
+size_t persistent_bwd_lds(int B, int H);
 void launch_lstm_persistent_bwd(const bf16* dY, const bf16* rec,
-                                const bf16* W_h_T, bf16* dG,
+                                const bf16* W_h_T, bf16* dG, bf16* dG_packT,
                                 unsigned int* pstate,
                                 unsigned int* abort_flag, int T, int B,
                                 int H, hipStream_t stream);

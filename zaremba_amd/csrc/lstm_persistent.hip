@@ -313,39 +313,41 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
 }
 
 // ===========================================================================
-// Backward
+// Backward (v2)
 // ===========================================================================
-// Per step t = T-1..0, each block (owning h-units [j0, j0+HS)):
-//   1. elementwise dgate math for its units from the block record +
-//      dY[t] + its register-carried (dc, dh_rec) state,
-//   2. publish its dG[t] columns (write-through pairs into the standard
-//      [T,B,4H] dG tensor — consumed in-launch AND later by the dW
-//      GEMMs),
-//   3. grid barrier,
-//   4. recurrent hop: dh_rec = dG[t] @ W_h for its HS output columns —
-//      W_h^T slice LDS-resident, dG staged chunk-wise (plain loads),
-//      4 waves split K, LDS reduction back to per-thread dh_rec.
-constexpr int BWD_CHUNK = 1024;  // K elements staged per chunk
-
+// One launch per layer reverse unroll. Per step t = T-1..0, each of the
+// NB resident blocks (owning h-units [j0, j0+HS)):
+//   1. dgate elementwise for its units from the 6-channel block record +
+//      dY[t] + register-carried (dc, dh_rec) state (rec/dY prefetched
+//      during the previous step's MFMA phase),
+//   2. publishes its dgates write-through, BOTH as standard [T,B,4H] dG
+//      (consumed later by the dW GEMMs) and in MFMA fragment-packed
+//      layout (consumed in-launch: contiguous 1 KB wave loads straight
+//      from L2 — a chunk-staged LDS variant with its 12 barriers/step
+//      measured 2x slower),
+//   3. grid barrier (same XCD-grouped counter barrier as forward),
+//   4. recurrent hop dh_rec = dG @ W_h for its HS output columns: packed
+//      global A-fragments x the LDS-resident W_h^T slice, 4 waves
+//      round-robin the K steps, LDS reduction back to per-thread state.
+// The per-step fallback pair re-reads the 18 MB W_h^T from HBM every
+// step; here it is read once.
 template <int MAXB>
 __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_bwd_kernel(
     const bf16* __restrict__ dY,     // [T, B, H]
     const bf16* __restrict__ rec,    // [T][NB][B][6][HS]
     const bf16* __restrict__ W_h_T,  // [H, 4H] transposed shadow
     bf16* __restrict__ dG,           // [T, B, 4H] out (write-through)
+    bf16* __restrict__ dG_packT,     // [T][KS2*2*64*8] zero-prefilled
     unsigned int* __restrict__ pstate,
     unsigned int* __restrict__ abort_flag,
     int T, int B, int H, int HS) {
   const int K = 4 * H;
-  const int CPAD = BWD_CHUNK + 8;
+  const int KS2 = (K + 31) / 32;
+  const int KWPAD = KS2 * 32 + 8;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* Ws = reinterpret_cast<bf16*>(smem);                // [HS][KWPAD]
-  // rows cover whole staged chunks (reads go to chunk-rounded k); the
-  // zero fill past K keeps 0*garbage out of the MFMA
-  const int KWPAD = ((K + BWD_CHUNK - 1) / BWD_CHUNK) * BWD_CHUNK + 8;
-  bf16* as = Ws + (int64_t)HS * KWPAD;                     // [B][CPAD]
-  float* gbuf = reinterpret_cast<float*>(as + (int64_t)B * CPAD);
-  bf16* dgbuf = reinterpret_cast<bf16*>(gbuf + 4 * MAXB * 16);  // [B][4][HS]
+  bf16* Ws = reinterpret_cast<bf16*>(smem);                 // [HS][KWPAD]
+  float* gbuf = reinterpret_cast<float*>(Ws + (int64_t)HS * KWPAD);
+  bf16* dgbuf = reinterpret_cast<bf16*>(gbuf + 4 * B * 16);  // [B][4][HS]
 
   const int NB = (H + HS - 1) / HS;
   const int grp = blockIdx.x & 7;
@@ -375,11 +377,6 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_bwd_kernel(
     }
     *reinterpret_cast<bf16x8*>(Ws + (int64_t)row * KWPAD + kv) = v;
   }
-  // zero the chunk-image K-tail rows once (tail beyond staged chunk len)
-  for (int idx = t_; idx < B * (CPAD - BWD_CHUNK); idx += PCELL_THREADS) {
-    const int b = idx / (CPAD - BWD_CHUNK);
-    as[(int64_t)b * CPAD + BWD_CHUNK + idx % (CPAD - BWD_CHUNK)] = (bf16)0.f;
-  }
 
   const int own_b = t_ / HS;
   const int own_jj = t_ % HS;
@@ -391,34 +388,44 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_bwd_kernel(
   const int wc = (lm < HS ? lm : HS - 1);
   const int fr0 = (l >> 4) * 4;
   const int lk = (l >> 4) * 8;
+  (void)a0r; (void)a1r; (void)lk;
+
+  // prefetch the first step's record/dY for the owning thread
+  float r_i = 0, r_f = 0, r_o = 0, r_n = 0, r_tc = 0, r_cp = 0, r_dy = 0;
+  if (own) {
+    const bf16* r =
+        rec + ((((int64_t)(T - 1) * NB + blockIdx.x) * B + own_b) * REC_CH) *
+                  HS;
+    r_i = bf2f(r[0 * HS + own_jj]);
+    r_f = bf2f(r[1 * HS + own_jj]);
+    r_o = bf2f(r[2 * HS + own_jj]);
+    r_n = bf2f(r[3 * HS + own_jj]);
+    r_tc = bf2f(r[4 * HS + own_jj]);
+    r_cp = bf2f(r[5 * HS + own_jj]);
+    r_dy = bf2f(dY[((int64_t)(T - 1) * B + own_b) * H + j0 + own_jj]);
+  }
 
   for (int t = T - 1; t >= 0; --t) {
-    // ---- 1. elementwise dgates for own units -----------------------------
-    __syncthreads();  // previous iteration's dgbuf/as consumers done
+    __syncthreads();  // previous step's dgbuf consumers done
+    // ---- 1. dgate elementwise for own units ------------------------------
     if (own) {
-      const int b = own_b, jj = own_jj;
-      const int j = j0 + jj;
-      const bf16* r =
-          rec + ((((int64_t)t * NB + blockIdx.x) * B + b) * REC_CH) * HS;
-      const float i_ = bf2f(r[0 * HS + jj]);
-      const float f_ = bf2f(r[1 * HS + jj]);
-      const float o_ = bf2f(r[2 * HS + jj]);
-      const float n_ = bf2f(r[3 * HS + jj]);
-      const float tc = bf2f(r[4 * HS + jj]);
-      const float cprev = bf2f(r[5 * HS + jj]);
-      const float dh = bf2f(dY[((int64_t)t * B + b) * H + j]) + dh_rec;
-      const float do_ = dh * tc;
-      const float dct = dc_reg + dh * o_ * (1.f - tc * tc);
-      dgbuf[(b * 4 + 0) * HS + jj] = f2bf(dct * n_ * i_ * (1.f - i_));
-      dgbuf[(b * 4 + 1) * HS + jj] = f2bf(dct * cprev * f_ * (1.f - f_));
-      dgbuf[(b * 4 + 2) * HS + jj] = f2bf(do_ * o_ * (1.f - o_));
-      dgbuf[(b * 4 + 3) * HS + jj] = f2bf(dct * i_ * (1.f - n_ * n_));
-      dc_reg = dct * f_;
+      const float dh = r_dy + dh_rec;
+      const float do_ = dh * r_tc;
+      const float dct = dc_reg + dh * r_o * (1.f - r_tc * r_tc);
+      dgbuf[(own_b * 4 + 0) * HS + own_jj] =
+          f2bf(dct * r_n * r_i * (1.f - r_i));
+      dgbuf[(own_b * 4 + 1) * HS + own_jj] =
+          f2bf(dct * r_cp * r_f * (1.f - r_f));
+      dgbuf[(own_b * 4 + 2) * HS + own_jj] = f2bf(do_ * r_o * (1.f - r_o));
+      dgbuf[(own_b * 4 + 3) * HS + own_jj] =
+          f2bf(dct * r_i * (1.f - r_n * r_n));
+      dc_reg = dct * r_f;
     }
     __syncthreads();
-    // ---- 2. publish dG[t] columns (paired write-through) -----------------
+    // ---- 2. publish dG[t]: standard + fragment-packed (write-through) ----
     {
       bf16* dst = dG + (int64_t)t * B * K;
+      bf16* pkt = dG_packT + (int64_t)t * KS2 * 2 * 64 * 8;
       for (int i = t_; i < B * 4 * HS / 2; i += PCELL_THREADS) {
         const int b = i / (4 * HS / 2);
         const int rem = i % (4 * HS / 2);
@@ -426,86 +433,104 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_bwd_kernel(
         const int jj = (rem % (HS / 2)) * 2;
         const int j = j0 + jj;
         if (j < H) {
-          store_pair_wt(dst + (int64_t)b * K + gg * H + j,
-                        dgbuf[(b * 4 + gg) * HS + jj],
-                        (j + 1 < H) ? dgbuf[(b * 4 + gg) * HS + jj + 1]
-                                    : (bf16)0.f);
+          const bf16 v0 = dgbuf[(b * 4 + gg) * HS + jj];
+          const bf16 v1 = (j + 1 < H) ? dgbuf[(b * 4 + gg) * HS + jj + 1]
+                                      : (bf16)0.f;
+          const int k = gg * H + j;
+          store_pair_wt(dst + (int64_t)b * K + k, v0, v1);
+          const int ks = k / 32, sub = k % 32;
+          const int pl = (b & 15) + 16 * (sub / 8);
+          store_pair_wt(
+              pkt + ((((int64_t)ks * 2 + b / 16) * 64 + pl) * 8 + sub % 8),
+              v0, v1);
         }
       }
     }
-    // ---- 3. grid barrier -------------------------------------------------
+    // ---- 3. prefetch next step's record/dY (overlaps barrier + MFMA) -----
+    if (own && t > 0) {
+      const bf16* r =
+          rec +
+          ((((int64_t)(t - 1) * NB + blockIdx.x) * B + own_b) * REC_CH) * HS;
+      r_i = bf2f(r[0 * HS + own_jj]);
+      r_f = bf2f(r[1 * HS + own_jj]);
+      r_o = bf2f(r[2 * HS + own_jj]);
+      r_n = bf2f(r[3 * HS + own_jj]);
+      r_tc = bf2f(r[4 * HS + own_jj]);
+      r_cp = bf2f(r[5 * HS + own_jj]);
+      r_dy = bf2f(dY[((int64_t)(t - 1) * B + own_b) * H + j0 + own_jj]);
+    }
+    // ---- 4. grid barrier --------------------------------------------------
     if (!xcd_grid_barrier(pstate, grp, nbg, ngroups, (unsigned int)(T - t),
                           abort_flag))
       return;
-    // ---- 4. recurrent hop: dh_rec = dG[t] @ W_h (own columns) ------------
+    // ---- 5. recurrent hop: packed global A x LDS W -----------------------
     f32x4 acc0 = {}, acc1 = {};
-    const bf16* dsrc = dG + (int64_t)t * B * K;
-    for (int k0 = 0; k0 < K; k0 += BWD_CHUNK) {
-      const int klen = min(BWD_CHUNK, K - k0);
-      // stage the chunk (plain loads; zero-fill tail inside the chunk)
-      for (int idx = t_; idx < B * (BWD_CHUNK / 8); idx += PCELL_THREADS) {
-        const int b = idx / (BWD_CHUNK / 8);
-        const int k = (idx % (BWD_CHUNK / 8)) * 8;
-        bf16x8 v = {};
-        if (k < klen) {
-          const bf16* p = dsrc + (int64_t)b * K + k0 + k;
-          if (k + 8 <= klen) {
-            v = *reinterpret_cast<const bf16x8*>(p);
-          } else {
+    {
+      const bf16x8* pa = reinterpret_cast<const bf16x8*>(
+                             dG_packT + (int64_t)t * KS2 * 2 * 64 * 8) +
+                         l;
+      const bf16* pw = Ws + (int64_t)wc * KWPAD;
+      const int nown = (KS2 - w + 3) / 4;  // owned steps: w + 4i
+      int i = 0;
+      for (; i + 8 <= nown; i += 8) {
+        bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
-            for (int e = 0; e < 8; ++e)
-              v[e] = (k + e < klen) ? p[e] : (bf16)0.f;
-          }
+        for (int u = 0; u < 8; ++u) {
+          const int ks = w + 4 * (i + u);
+          a0v[u] = pa[ks * 128];
+          a1v[u] = pa[ks * 128 + 64];
+          bwv[u] = *reinterpret_cast<const bf16x8*>(pw + ks * 32 +
+                                                    ((l >> 4) * 8));
         }
-        *reinterpret_cast<bf16x8*>(as + (int64_t)b * CPAD + k) = v;
-      }
-      __syncthreads();
-      // waves split the chunk: wave w owns k in [w*256, w*256+256)
-      const bf16* pw = Ws + (int64_t)wc * KWPAD + k0;
-      const bf16* pa0 = as + (int64_t)a0r * CPAD;
-      const bf16* pa1 = as + (int64_t)a1r * CPAD;
-      const int kw0 = w * (BWD_CHUNK / 4);
 #pragma unroll
-      for (int ks = 0; ks < BWD_CHUNK / 4; ks += 32) {
-        const int k = kw0 + ks + lk;
-        bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + k);
-        bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + k);
-        bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pw + k);
+        for (int u = 0; u < 8; ++u) {
+          acc0 = mfma_16x16x32_bf16(a0v[u], bwv[u], acc0);
+          acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
+        }
+      }
+      for (; i < nown; ++i) {
+        const int ks = w + 4 * i;
+        bf16x8 a0v = pa[ks * 128];
+        bf16x8 a1v = pa[ks * 128 + 64];
+        bf16x8 bwv = *reinterpret_cast<const bf16x8*>(pw + ks * 32 +
+                                                      ((l >> 4) * 8));
         acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
         acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
       }
-      __syncthreads();
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      gbuf[(w * MAXB + fr0 + r) * 16 + lm] = acc0[r];
-      gbuf[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+      if (fr0 + r < B) gbuf[(w * B + fr0 + r) * 16 + lm] = acc0[r];
+      if (16 + fr0 + r < B) gbuf[(w * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
     }
     __syncthreads();
     if (own) {
-      const int b = own_b, jj = own_jj;
-      dh_rec = gbuf[(0 * MAXB + b) * 16 + jj] +
-               gbuf[(1 * MAXB + b) * 16 + jj] +
-               gbuf[(2 * MAXB + b) * 16 + jj] +
-               gbuf[(3 * MAXB + b) * 16 + jj];
+      dh_rec = gbuf[(0 * B + own_b) * 16 + own_jj] +
+               gbuf[(1 * B + own_b) * 16 + own_jj] +
+               gbuf[(2 * B + own_b) * 16 + own_jj] +
+               gbuf[(3 * B + own_b) * 16 + own_jj];
     }
   }
 }
 
+size_t persistent_bwd_lds(int B, int H) {
+  const int HS = persistent_hs(H);
+  const int KWPAD = ((4 * H + 31) / 32) * 32 + 8;
+  return (size_t)HS * KWPAD * 2 + (size_t)4 * B * 16 * 4 +
+         (size_t)B * 4 * HS * 2 + 16;
+}
+
 void launch_lstm_persistent_bwd(const bf16* dY, const bf16* rec,
-                                const bf16* W_h_T, bf16* dG,
+                                const bf16* W_h_T, bf16* dG, bf16* dG_packT,
                                 unsigned int* pstate,
                                 unsigned int* abort_flag, int T, int B,
                                 int H, hipStream_t stream) {
   const int HS = persistent_hs(H);
   const int NB = cdiv(H, HS);
-  const int K = 4 * H;
-  const int KWPAD = ((K + BWD_CHUNK - 1) / BWD_CHUNK) * BWD_CHUNK + 8;
-  size_t lds = (size_t)HS * KWPAD * 2 + (size_t)B * (BWD_CHUNK + 8) * 2 +
-               4 * 32 * 16 * sizeof(float) + (size_t)B * 4 * HS * 2 + 16;
   hipLaunchKernelGGL((lstm_persistent_bwd_kernel<32>), dim3(NB),
-                     dim3(PCELL_THREADS), lds, stream, dY, rec, W_h_T, dG,
-                     pstate, abort_flag, T, B, H, HS);
+                     dim3(PCELL_THREADS), persistent_bwd_lds(B, H), stream,
+                     dY, rec, W_h_T, dG, dG_packT, pstate, abort_flag, T, B,
+                     H, HS);
 }
 
 }  // namespace zamd

@@ -44,7 +44,10 @@ class _LayerWorkspace:
         # must read as 0.0 in the packed cell kernels)
         self.h_pack = torch.zeros(T + 1, _ks(H) * 2 * 64 * 8, dtype=bf,
                                   device=device)
-        self.dG_pack = torch.zeros(_ks(4 * H) * 2 * 64 * 8, dtype=bf,
+        # per-t packed dgates: the persistent backward's in-launch
+        # exchange buffers (fresh slot per step); the per-step fallback
+        # reuses slot 0
+        self.dG_pack = torch.zeros(T, _ks(4 * H) * 2 * 64 * 8, dtype=bf,
                                    device=device)
         # persistent-kernel state: block records [T][NB][B][6][HS],
         # barrier words (re-zeroed per call by the driver) + abort flag
