@@ -31,39 +31,46 @@ __global__ void census_kernel(unsigned int* cnt, unsigned int* result,
 }
 
 // the same two-level barrier as lstm_persistent.hip, iterated
+template <int SLEEP, int VARIANT>  // VARIANT 0: two-level; 1: flat top-counter poll; 2: two-level no-acquire (timing only)
 __device__ bool xcd_barrier(unsigned int* pstate, int grp, int nbg,
-                            int ngroups, unsigned int gen,
+                            int ngroups, int nb, unsigned int gen,
                             unsigned int* fail) {
   __shared__ int ok_s;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   if (threadIdx.x == 0) {
     ok_s = 1;
     gu32* st = (gu32*)(uintptr_t)pstate;
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    unsigned int t = __hip_atomic_fetch_add(&st[grp], 1u, RLX_AGENT);
-    if (t == gen * nbg - 1) {
-      unsigned int tt = __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
-      if (tt == gen * ngroups - 1) {
-        for (int x = 0; x < 8; ++x)
-          __hip_atomic_store(&st[9 + x], gen, RLX_AGENT);
+    if (VARIANT == 1) {
+      __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
+      unsigned int spins = 0;
+      while (__hip_atomic_load(&st[8], RLX_AGENT) < gen * nb) {
+        __builtin_amdgcn_s_sleep(SLEEP);
+        if (++spins > 3000000u) { atomicAdd(fail, 1u); ok_s = 0; break; }
       }
-    }
-    unsigned int spins = 0;
-    while (__hip_atomic_load(&st[9 + grp], RLX_AGENT) < gen) {
-      __builtin_amdgcn_s_sleep(8);
-      if (++spins > 3000000u) {
-        atomicAdd(fail, 1u);
-        ok_s = 0;
-        break;
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    } else {
+      unsigned int t = __hip_atomic_fetch_add(&st[grp], 1u, RLX_AGENT);
+      if (t == gen * nbg - 1) {
+        unsigned int tt = __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
+        if (tt == gen * ngroups - 1) {
+          for (int x = 0; x < 8; ++x)
+            __hip_atomic_store(&st[9 + x], gen, RLX_AGENT);
+        }
       }
+      unsigned int spins = 0;
+      while (__hip_atomic_load(&st[9 + grp], RLX_AGENT) < gen) {
+        __builtin_amdgcn_s_sleep(SLEEP);
+        if (++spins > 3000000u) { atomicAdd(fail, 1u); ok_s = 0; break; }
+      }
+      if (VARIANT != 2) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   __syncthreads();
   return ok_s != 0;
 }
 
+template <int SLEEP, int VARIANT>
 __global__ void barrier_probe_kernel(unsigned int* pstate,
                                      unsigned int* fail, unsigned int* done,
                                      int iters, int nb) {
@@ -73,7 +80,8 @@ __global__ void barrier_probe_kernel(unsigned int* pstate,
   const int ngroups = nb < 8 ? nb : 8;
   const int nbg = (nb - grp + 7) / 8;
   for (int t = 1; t <= iters; ++t) {
-    if (!xcd_barrier(pstate, grp, nbg, ngroups, (unsigned int)t, fail))
+    if (!xcd_barrier<SLEEP, VARIANT>(pstate, grp, nbg, ngroups, nb,
+                                     (unsigned int)t, fail))
       return;
   }
   if (threadIdx.x == 0) atomicAdd(done, 1u);
@@ -93,25 +101,30 @@ int main() {
     (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
     printf("census nb=%3d: full=%u timeout=%u\n", nb, res[0], res[1]);
   }
-  // barrier probe at the largest resident size (timed)
-  for (int nb : {250, 125}) {
-    for (int iters : {1000}) {
+  // barrier probe at the persistent-kernel sizes (timed)
+  for (int nb : {188, 125, 94}) {
+    auto run = [&](const char* name, auto kern) {
       (void)hipMemset(buf, 0, 4096);
       hipEvent_t e0, e1;
       (void)hipEventCreate(&e0);
       (void)hipEventCreate(&e1);
       (void)hipEventRecord(e0, 0);
-      hipLaunchKernelGGL(barrier_probe_kernel, dim3(nb), dim3(256), lds, 0,
-                         buf, buf + 64, buf + 65, iters, nb);
+      hipLaunchKernelGGL(kern, dim3(nb), dim3(256), lds, 0, buf, buf + 64,
+                         buf + 65, 1000, nb);
       (void)hipEventRecord(e1, 0);
       (void)hipDeviceSynchronize();
       float ms = 0;
       (void)hipEventElapsedTime(&ms, e0, e1);
       unsigned int res[2];
       (void)hipMemcpy(res, buf + 64, 8, hipMemcpyDeviceToHost);
-      printf("barrier nb=%3d iters=%d: fail=%u done=%u  %.2f us/barrier\n",
-             nb, iters, res[0], res[1], ms * 1000.f / iters);
-    }
+      printf("barrier nb=%3d %-22s: fail=%u done=%u  %.2f us/barrier\n", nb,
+             name, res[0], res[1], ms * 1000.f / 1000);
+    };
+    run("2lvl sleep8", barrier_probe_kernel<8, 0>);
+    run("2lvl sleep2", barrier_probe_kernel<2, 0>);
+    run("2lvl sleep32", barrier_probe_kernel<32, 0>);
+    run("flat sleep8", barrier_probe_kernel<8, 1>);
+    run("2lvl sleep8 noacq", barrier_probe_kernel<8, 2>);
   }
   return 0;
 }
