@@ -153,30 +153,55 @@ void launch_dropout_bwd(const bf16* dy, bf16* dx, float p, uint64_t seed,
 // sum_n (lse_n - s_n[y_n]) into loss_accum (zeroed by the host wrapper);
 // saves lse for backward. Loss scale (batch_size / N) applied host-side
 // as a lazy torch op.
+// single pass: per-thread online (max, sum) over its stripe, then a
+// wave/block reduction with the online-softmax combine.
 __global__ void lsm_nll_fwd_kernel(const float* __restrict__ scores,
                                    const int64_t* __restrict__ y,
                                    float* __restrict__ lse,
                                    float* __restrict__ loss_accum, int N,
                                    int V) {
-  __shared__ float scratch[8];
+  __shared__ float sm[8], ss[8];
   int row = blockIdx.x;
   if (row >= N) return;
   const float* s = scores + (int64_t)row * V;
-  float m = -INFINITY;
-  for (int v = threadIdx.x; v < V; v += blockDim.x) m = fmaxf(m, s[v]);
-  m = block_reduce(m, scratch, -INFINITY,
-                   [] __device__(float a, float b) { return fmaxf(a, b); });
+  float m = -INFINITY, acc = 0.f;
+  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+    float x = s[v];
+    if (x > m) {
+      acc = acc * __expf(m - x) + 1.f;
+      m = x;
+    } else {
+      acc += __expf(x - m);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float m2 = __shfl_down(m, off, 64);
+    float a2 = __shfl_down(acc, off, 64);
+    float mn = fmaxf(m, m2);
+    // guard the both-(-inf) case (threads whose stripe was empty):
+    // exp(-inf - -inf) would be NaN
+    float pa = (m == -INFINITY) ? 0.f : acc * __expf(m - mn);
+    float pb = (m2 == -INFINITY) ? 0.f : a2 * __expf(m2 - mn);
+    acc = pa + pb;
+    m = mn;
+  }
+  const int nw = blockDim.x / 64;
+  if (lane_id() == 0) {
+    sm[wave_id()] = m;
+    ss[wave_id()] = acc;
+  }
   __syncthreads();
-  if (threadIdx.x == 0) scratch[0] = m;
-  __syncthreads();
-  m = scratch[0];
-  float acc = 0.f;
-  for (int v = threadIdx.x; v < V; v += blockDim.x) acc += __expf(s[v] - m);
-  __syncthreads();
-  acc = block_reduce(acc, scratch, 0.f,
-                     [] __device__(float a, float b) { return a + b; });
   if (threadIdx.x == 0) {
-    float l = m + __logf(acc);
+    float M = sm[0], A = ss[0];
+    for (int w = 1; w < nw; ++w) {
+      float mn = fmaxf(M, sm[w]);
+      float pa = (M == -INFINITY) ? 0.f : A * __expf(M - mn);
+      float pb = (sm[w] == -INFINITY) ? 0.f : ss[w] * __expf(sm[w] - mn);
+      A = pa + pb;
+      M = mn;
+    }
+    float l = M + __logf(A);
     lse[row] = l;
     atomicAdd(loss_accum, l - s[y[row]]);
   }

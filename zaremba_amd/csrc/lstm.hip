@@ -237,9 +237,10 @@ __global__ void lstm_cell_bwd_elt_kernel(
   float n_ = bf2f(r[3 * HSp + jr]);
   float tc = bf2f(r[4 * HSp + jr]);
   float cprev = bf2f(r[5 * HSp + jr]);
+  // first reverse step (dh_rec == null): carried state starts at zero
   float dh = bf2f(dy[idx]) + (dh_rec ? dh_rec[idx] + dh_rec2[idx] : 0.f);
   float do_ = dh * tc;
-  float dct = dc[idx] + dh * o_ * (1.f - tc * tc);
+  float dct = (dh_rec ? dc[idx] : 0.f) + dh * o_ * (1.f - tc * tc);
   float di = dct * n_;
   float df = dct * cprev;
   float dn = dct * i_;

@@ -180,7 +180,6 @@ class LstmLayerFn(torch.autograd.Function):
         T, B, H = ws.dY.shape
         Hin = x2.size(2)
         ws.dY.copy_(dY.to(torch.bfloat16))
-        ws.dc.zero_()
         e.lstm_seq_bwd(ws.dY, ws.gates, ws.rec, ws.c_all, rt.WhT, rt.WhTP,
                        ws.dG, ws.dG_pack, ws.dh_rec, ws.dc, ws.hgran,
                        ws.abort)
