@@ -98,16 +98,21 @@ def setdevice(args, plural=False):
 
 def load_data(args):
     if args.data.startswith("synthetic"):
-        opts = {"vocab": 10000, "branch": 20}
+        opts = {"vocab": 10000, "branch": 20, "tokens": 929589}
         if ":" in args.data:
             for kv in args.data.split(":", 1)[1].split(","):
                 k, v = kv.split("=")
                 opts[k] = int(v)
         seed = args.seed if args.seed is not None else 1234
+        sizes = dict(train_tokens=opts["tokens"],
+                     valid_tokens=max(400, opts["tokens"] // 12),
+                     test_tokens=max(400, opts["tokens"] // 11))
         if args.data.startswith("synthetic_markov"):
             return zdata.synthetic_markov_init(
-                vocab_size=opts["vocab"], branch=opts["branch"], seed=seed)
-        return zdata.synthetic_init(vocab_size=opts["vocab"], seed=seed)
+                vocab_size=opts["vocab"], branch=opts["branch"], seed=seed,
+                **sizes)
+        return zdata.synthetic_init(vocab_size=opts["vocab"], seed=seed,
+                                    **sizes)
     return zdata.data_init(args.data_dir)
 
 

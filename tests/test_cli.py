@@ -1,0 +1,62 @@
+"""CLI contract tests: main.py / ensemble.py run end-to-end on CPU with
+the reference flag surface (tiny synthetic configs)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_cli(args, timeout=300):
+    return subprocess.run([sys.executable] + args, cwd=REPO, timeout=timeout,
+                          capture_output=True, text=True)
+
+
+@pytest.mark.timeout(300)
+def test_main_cli_cpu_tiny(tmp_path):
+    ck = str(tmp_path / "m.pt")
+    jl = str(tmp_path / "m.jsonl")
+    r = run_cli(["main.py", "--device", "cpu", "--data",
+                 "synthetic:vocab=50,tokens=6000", "--hidden_size", "32",
+                 "--layer_num", "2", "--seq_length", "10", "--batch_size",
+                 "4", "--total_epochs", "2", "--dropout", "0.2", "--seed",
+                 "1", "--save", ck, "--jsonl", jl, "--lstm_type", "custom"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "Model will be training on the CPU." in r.stdout
+    assert "Validation set perplexity" in r.stdout
+    assert "Test set perplexity" in r.stdout
+    assert "Training is over." in r.stdout
+    assert os.path.exists(ck)
+    events = [json.loads(l) for l in open(jl)]
+    assert any(e["event"] == "final" for e in events)
+    # resume from the checkpoint (epoch 2 of 3)
+    r2 = run_cli(["main.py", "--device", "cpu", "--data",
+                  "synthetic:vocab=50,tokens=6000", "--hidden_size", "32",
+                  "--layer_num", "2", "--seq_length", "10", "--batch_size",
+                  "4", "--total_epochs", "3", "--seed", "1",
+                  "--resume", ck])
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert "Resumed from" in r2.stdout
+
+
+@pytest.mark.timeout(300)
+def test_ensemble_cli_cpu_tiny(tmp_path):
+    r = run_cli(["ensemble.py", "--device", "cpu", "--data",
+                 "synthetic:vocab=40,tokens=4000", "--hidden_size", "24",
+                 "--layer_num", "1", "--seq_length", "8", "--batch_size",
+                 "4", "--total_epochs", "1", "--ensemble_num", "2",
+                 "--seed", "2", "--save_dir", str(tmp_path)])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "Test set perplexity of 2 averaged models" in r.stdout
+    assert os.path.exists(str(tmp_path / "model_2.pt"))
+
+
+@pytest.mark.timeout(120)
+def test_main_cli_missing_ptb_message(tmp_path):
+    r = run_cli(["main.py", "--device", "cpu", "--data_dir", str(tmp_path)])
+    assert r.returncode != 0
+    assert "synthetic" in (r.stderr + r.stdout)
