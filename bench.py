@@ -51,6 +51,8 @@ def main():
     from zaremba_amd.models.lstm_lm import Model
     from zaremba_amd.parallel import dist as zdist
 
+    if args.engine == "eager":
+        os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
     rank, world = zdist.rank(), zdist.world_size()
     has_gpu = torch.cuda.is_available()

@@ -88,8 +88,13 @@ def load_data(args):
 def train_one(args, vocab_size, data, model_num):
     if args.seed is not None:
         torch.manual_seed(args.seed + 1000 * model_num)
+    engine = args.engine
+    if args.device.type == "cuda" and args.dtype == "fp32" \
+            and engine == "auto":
+        engine = "eager"
+        os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     model = Model(vocab_size, args.hidden_size, args.layer_num, args.dropout,
-                  args.winit, args.lstm_type, engine=args.engine)
+                  args.winit, args.lstm_type, engine=engine)
     model.to(args.device)
     if args.device.type == "cuda" and args.dtype == "bf16":
         model.hip().set_compute_dtype(torch.bfloat16)

@@ -147,6 +147,12 @@ def main():
         model = Model(vocab_size, args.hidden_size, args.layer_num,
                       args.dropout, args.winit, args.lstm_type,
                       engine=args.engine)
+    if args.device.type == "cuda" and args.dtype == "fp32" \
+            and args.engine == "auto":
+        # fp32 compute on GPU = the eager engine (the HIP kernel path is
+        # bf16-compute / fp32-master by design)
+        model.engine = "eager"
+        os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     model.to(args.device)
     if args.device.type == "cuda" and args.dtype == "bf16":
         model.hip().set_compute_dtype(torch.bfloat16)
