@@ -21,8 +21,6 @@ MI355X-native specifics:
 
 from __future__ import annotations
 
-from typing import List, Tuple
-
 import torch
 import torch.nn as nn
 
@@ -145,11 +143,20 @@ class Model(nn.Module):
 
     def hip(self):
         """Return (building if needed) the HIP execution plan for this model."""
+        dev = next(self.parameters()).device
+        if self._hip_model is not None and self._hip_model.device != dev:
+            self._hip_model = None  # model moved devices: rebuild the plan
         if self._hip_model is None:
             from ..ops.hip_model import HipModel
 
             self._hip_model = HipModel(self)
         return self._hip_model
+
+    def load_state_dict(self, *a, **kw):
+        out = super().load_state_dict(*a, **kw)
+        if self._hip_model is not None:
+            self._hip_model.invalidate_shadows()
+        return out
 
     # -- forward (reference model.py:103-110) --------------------------------
 

@@ -16,7 +16,7 @@ Owns, per model instance:
 from __future__ import annotations
 
 import secrets
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
