@@ -1,0 +1,35 @@
+import sys, os, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+import zaremba_amd._hip as ext
+
+dev = "cuda"
+def bench(M, N, K, out_dtype=torch.float32, iters=30):
+    A = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    B = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+    C = torch.empty(M, N, device=dev, dtype=out_dtype)
+    for _ in range(5): ext.gemm(A, B, C, None, False, False)
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(iters): ext.gemm(A, B, C, None, False, False)
+    torch.cuda.synchronize(); dt = (time.perf_counter()-t0)/iters
+    tf = 2*M*N*K/dt/1e12
+    print(f"NT M={M:6d} N={N:6d} K={K:6d} {str(out_dtype)[6:]:9s}: "
+          f"{dt*1e6:8.1f} us  {tf:7.1f} TF")
+
+# dW shapes: measured in-train ~56us
+bench(6000, 1500, 700)
+bench(6000, 1500, 704)   # no K-tail, 11 tiles
+bench(6016, 1536, 768)   # all-interior variant
+bench(6000, 1500, 768)   # K padded only
+bench(10000, 1500, 700)
+bench(10048, 1536, 768)
+# dx shapes (bf16 out): in-train ~88us
+bench(700, 1500, 6000, torch.bfloat16)
+bench(768, 1536, 6016, torch.bfloat16)
+bench(700, 1500, 10000, torch.bfloat16)
+# input gemm shapes
+bench(700, 6000, 1500, torch.bfloat16)
+bench(768, 6016, 1536, torch.bfloat16)
+# proj
+bench(700, 10000, 1500)
+bench(768, 10048, 1536)

@@ -388,7 +388,8 @@ void launch_sgd_update(float* master, const float* grad, bf16* shadow,
 // 64x64 LDS tile, 16-B vector loads AND stores; interior blocks take a
 // guard-free path (per-element guards around global loads serialize).
 __global__ void transpose_bf16_kernel(const bf16* __restrict__ src,
-                                      bf16* __restrict__ dst, int R, int C) {
+                                      bf16* __restrict__ dst, int R, int C,
+                                      int ldd) {  // dst row stride (>= R)
   __shared__ bf16 tile[64][72];  // +8 bf16 row pad (16 B): conflict relief
   const int c0 = blockIdx.x * 64, r0 = blockIdx.y * 64;
   const int t = threadIdx.x;
@@ -424,20 +425,21 @@ __global__ void transpose_bf16_kernel(const bf16* __restrict__ src,
 #pragma unroll
     for (int e = 0; e < 8; ++e) v[e] = tile[lc8 + e][c];
     if (interior) {
-      *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + c) * R + r0 + lc8) = v;
+      *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + c) * ldd + r0 + lc8) = v;
     } else if (c0 + c < C) {
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        if (r0 + lc8 + e < R) dst[(int64_t)(c0 + c) * R + r0 + lc8 + e] = v[e];
+        if (r0 + lc8 + e < R)
+          dst[(int64_t)(c0 + c) * ldd + r0 + lc8 + e] = v[e];
     }
   }
 }
 
 void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
-                           hipStream_t stream) {
+                           int ldd, hipStream_t stream) {
   dim3 grid(cdiv(C, 64), cdiv(R, 64));
   hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream, src,
-                     dst, R, C);
+                     dst, R, C, ldd);
 }
 
 // column sum of a bf16 matrix -> f32 (bias grads: db = colsum(dG)).

@@ -414,10 +414,14 @@ static void colsum_bf16(const torch::Tensor& in, torch::Tensor& out) {
   launch_colsum_bf16(bf_ptr(in), f_ptr_mut(out), R, C, current_stream());
 }
 
+// dst is [C, ldd] with ldd >= R; columns [R, ldd) are left untouched
+// (pre-zero them once when using the pad as a GEMM K extension).
 static void transpose_bf16(const torch::Tensor& src, torch::Tensor& dst) {
   int R = src.size(0), C = src.size(1);
-  TORCH_CHECK(dst.size(0) == C && dst.size(1) == R);
-  launch_transpose_bf16(bf_ptr(src), bf_ptr_mut(dst), R, C, current_stream());
+  int ldd = dst.size(1);
+  TORCH_CHECK(dst.size(0) == C && ldd >= R);
+  launch_transpose_bf16(bf_ptr(src), bf_ptr_mut(dst), R, C, ldd,
+                        current_stream());
 }
 
 }  // namespace zamd

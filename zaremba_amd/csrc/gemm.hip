@@ -44,10 +44,15 @@ DEV_INLINE void glds16(const bf16* gsrc, bf16* lds_dst_uniform) {
 }
 
 // Stage a [TILE][BK] k-contiguous tile via glds: TILE*BK*2/1024
-// instructions split across 4 waves.
+// instructions split across 4 waves. Out-of-range rows are CLAMPED to
+// the last valid row: the garbage values land only in output rows/cols
+// the epilogue discards (each output element depends only on its own
+// A row and B col), so M/N-edge blocks ride the same fast path — only
+// K-tail tiles need the guarded register staging (tail columns feed
+// REAL outputs and must be zero).
 template <int TILE>
 DEV_INLINE void stage_glds_kc(const bf16* __restrict__ src, int ld, int row0,
-                              int k0, bf16* lds) {
+                              int nrows_total, int k0, bf16* lds) {
   constexpr int NINST = TILE * BK * 2 / 1024;     // 16 (TILE=128) or 4
   constexpr int PER_WAVE = NINST / 4;
   const int w = wave_id();
@@ -56,9 +61,10 @@ DEV_INLINE void stage_glds_kc(const bf16* __restrict__ src, int ld, int row0,
   for (int i = 0; i < PER_WAVE; ++i) {
     const int inst = w * PER_WAVE + i;
     const int row = inst * 8 + (l >> 3);          // 8 rows per instruction
+    const int gr = min(row0 + row, nrows_total - 1);
     const int colb = (l & 7) * 16;                // byte column 0..112
     const int src_colb = colb ^ ((row & 7) << 4); // pre-swizzled source
-    glds16(src + (int64_t)(row0 + row) * ld + k0 + src_colb / 2,
+    glds16(src + (int64_t)gr * ld + k0 + src_colb / 2,
            lds + (int64_t)inst * 512);            // 1024 B = 512 bf16
   }
 }
@@ -76,18 +82,21 @@ DEV_INLINE void stage_load_kc(const bf16* __restrict__ src, int ld, int row0,
   for (int p = 0; p < BROWS / 32; ++p) {
     const int row = r + p * 32;
     if (GUARD) {
-      v[p] = bf16x8{};
-      const int gr = row0 + row;
+      // Branch-free guards: loads always execute from CLAMPED in-bounds
+      // addresses; out-of-range values are zeroed by VALUE selects.
+      // (Branching around loads makes hipcc drain vmcnt per element —
+      // measured ~12 us per K-tail tile before.) Row clamp alone is
+      // enough for the M/N direction: pad-row garbage only reaches
+      // discarded outputs; the K direction must read as ZERO.
+      // all-scalar, all-unconditional: even a per-thread branch around a
+      // vector load serializes hipcc's waitcnt bookkeeping
+      const int gr = row0 + min(row, nrows - 1);
       const int gk = k0 + kk;
-      if (row < nrows) {
-        const bf16* sp = src + (int64_t)gr * ld + gk;
-        if (gk + 8 <= K) {
-          v[p] = *reinterpret_cast<const bf16x8*>(sp);
-        } else {
+      const bf16* sp = src + (int64_t)gr * ld;
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            v[p][e] = (gk + e < K) ? sp[e] : (bf16)0.f;
-        }
+      for (int e = 0; e < 8; ++e) {
+        bf16 x = sp[min(gk + e, K - 1)];
+        v[p][e] = (gk + e < K) ? x : (bf16)0.f;
       }
     } else {
       v[p] = *reinterpret_cast<const bf16x8*>(src + (int64_t)(row0 + row) * ld +
@@ -121,13 +130,13 @@ DEV_INLINE void stage_load_tr(const bf16* __restrict__ src, int ld, int row0,
   for (int p = 0; p < BROWS / 32; ++p) {    // BK/KG passes == BROWS/32
     const int kk = kb + p * KG;
     if (GUARD) {
-      v[p] = bf16x8{};
-      if (row < nrows) {
+      // clamped-address loads + value selects (see stage_load_kc note)
+      const int grc = row0 + min(row, nrows - 1);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int gk = k0 + kk + e;
-          if (gk < K) v[p][e] = src[(int64_t)gk * ld + gr];
-        }
+      for (int e = 0; e < 8; ++e) {
+        const int gk = k0 + kk + e;
+        bf16 x = src[(int64_t)min(gk, K - 1) * ld + grc];
+        v[p][e] = (gk < K) ? x : (bf16)0.f;
       }
     } else {
 #pragma unroll
@@ -236,18 +245,19 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     }
   };
 
-  if (!TRANS_A && !TRANS_B && interior && k_full > 0) {
+  if (!TRANS_A && !TRANS_B && k_full > 0) {
     // glds 2-phase pipeline (guide §5.5 T3 minimum form): stage tile t+1
     // while computing tile t; __syncthreads() drains the in-flight glds.
-    stage_glds_kc<TILE>(A, lda, m0, 0, As);
-    stage_glds_kc<TILE>(B, ldb, n0, 0, Bs);
+    // Edge blocks use row-clamped sources (see stage_glds_kc).
+    stage_glds_kc<TILE>(A, lda, m0, M, 0, As);
+    stage_glds_kc<TILE>(B, ldb, n0, N, 0, Bs);
     __syncthreads();
     int cur = 0;
     for (int kt = 0; kt < k_full; ++kt) {
       if (kt + 1 < k_full) {
-        stage_glds_kc<TILE>(A, lda, m0, (kt + 1) * BK,
+        stage_glds_kc<TILE>(A, lda, m0, M, (kt + 1) * BK,
                             As + (cur ^ 1) * TILE * BK);
-        stage_glds_kc<TILE>(B, ldb, n0, (kt + 1) * BK,
+        stage_glds_kc<TILE>(B, ldb, n0, N, (kt + 1) * BK,
                             Bs + (cur ^ 1) * TILE * BK);
       }
       mfma_phase(As + cur * TILE * BK, Bs + cur * TILE * BK);
@@ -255,8 +265,10 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
       cur ^= 1;
     }
     if (k_full < nk) {  // K tail: register-staged, guarded
-      stage_load_kc<TILE, true>(A, lda, m0, TILE, k_full * BK, K, va);
-      stage_load_kc<TILE, true>(B, ldb, n0, TILE, k_full * BK, K, vb);
+      stage_load_kc<TILE, true>(A, lda, m0, min(TILE, M - m0), k_full * BK,
+                                K, va);
+      stage_load_kc<TILE, true>(B, ldb, n0, min(TILE, N - n0), k_full * BK,
+                                K, vb);
       stage_write_kc<TILE>(As + cur * TILE * BK, va);
       stage_write_kc<TILE>(Bs + cur * TILE * BK, vb);
       __syncthreads();

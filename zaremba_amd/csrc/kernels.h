@@ -76,7 +76,7 @@ void launch_sgd_update(float* master, const float* grad, bf16* shadow,
                        const float* norm2, float max_norm, float lr,
                        float grad_scale, int64_t n, hipStream_t stream);
 void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
-                           hipStream_t stream);
+                           int ldd, hipStream_t stream);
 void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                         hipStream_t stream);
 

@@ -49,6 +49,13 @@ class _LayerWorkspace:
         # reuses slot 0
         self.dG_pack = torch.zeros(T, _ks(4 * H) * 2 * 64 * 8, dtype=bf,
                                    device=device)
+        # zero-padded transposed temporaries for the dW GEMMs: K (=T*B)
+        # padded to a BK multiple so no GEMM runs a K-tail tile; pad
+        # columns stay zero (transpose writes only [:, :T*B])
+        TBp = ((T * B + 63) // 64) * 64
+        self.TBp = TBp
+        self.dGT = torch.zeros(4 * H, TBp, dtype=bf, device=device)
+        self.hpT = torch.zeros(H, TBp, dtype=bf, device=device)
         # persistent-kernel state: block records [T][NB][B][6][HS],
         # barrier words (re-zeroed per call by the driver) + abort flag
         hs = _C.ext().persistent_hs(H)
@@ -102,6 +109,8 @@ class _FcRuntime:
         V, H = fc.W.shape
         self.W = torch.empty(V, H, dtype=bf, device=device)
         self.WT = torch.empty(H, V, dtype=bf, device=device)
+        self.dscT = None  # [V, TBp] zero-padded, sized on first backward
+        self.xT = None    # [H, TBp]
 
     @torch.no_grad()
     def refresh(self, fc, e):

@@ -97,13 +97,19 @@ class LinearFn(torch.autograd.Function):
         dx = torch.empty(N, H, dtype=torch.bfloat16, device=x.device)
         e.gemm(dsc, fc_rt.WT, dx, None, False, False)
         # dW = dsc^T @ x via explicit transposes + the fast NT kernel (the
-        # TN staging path is register-starved; measured 2-3x slower)
-        dscT = torch.empty(V, N, dtype=torch.bfloat16, device=x.device)
-        e.transpose_bf16(dsc, dscT)
-        xT = torch.empty(H, N, dtype=torch.bfloat16, device=x.device)
-        e.transpose_bf16(x, xT)
+        # TN staging path is register-starved; measured 2-3x slower).
+        # Transposed temps are zero-padded in K so the GEMM has no K-tail
+        # tile (a tail tile measured +13.5 us).
+        Np = ((N + 63) // 64) * 64
+        if fc_rt.dscT is None or fc_rt.dscT.size(1) != Np:
+            fc_rt.dscT = torch.zeros(V, Np, dtype=torch.bfloat16,
+                                     device=x.device)
+            fc_rt.xT = torch.zeros(H, Np, dtype=torch.bfloat16,
+                                   device=x.device)
+        e.transpose_bf16(dsc, fc_rt.dscT)
+        e.transpose_bf16(x, fc_rt.xT)
         dW = torch.empty(V, H, dtype=torch.float32, device=x.device)
-        e.gemm(dscT, xT, dW, None, False, False)
+        e.gemm(fc_rt.dscT, fc_rt.xT, dW, None, False, False)
         db = dscores.sum(0)
         return dx, dW, db, None
 
@@ -187,13 +193,16 @@ class LstmLayerFn(torch.autograd.Function):
         dG2 = ws.dG.view(TB, 4 * H)
         # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x —
         # weight grads via explicit transposes + the fast NT kernel.
-        dGT = torch.empty(4 * H, TB, dtype=torch.bfloat16, device=x2.device)
+        # K (=T*B) is zero-padded in the persistent transposed temps so
+        # no dW GEMM runs a K-tail tile.
+        dGT = ws.dGT
         e.transpose_bf16(dG2, dGT)
-        hpT = torch.empty(H, TB, dtype=torch.bfloat16, device=x2.device)
+        hpT = ws.hpT
         e.transpose_bf16(ws.h_all[:T].reshape(TB, H), hpT)
         dWh = torch.empty(4 * H, H, dtype=torch.float32, device=x2.device)
         e.gemm(dGT, hpT, dWh, None, False, False)
-        xT = torch.empty(Hin, TB, dtype=torch.bfloat16, device=x2.device)
+        xT = torch.empty(Hin, ws.TBp, dtype=torch.bfloat16, device=x2.device)
+        xT[:, TB:].zero_()
         e.transpose_bf16(x2.view(TB, Hin), xT)
         dWx = torch.empty(4 * H, Hin, dtype=torch.float32, device=x2.device)
         e.gemm(dGT, xT, dWx, None, False, False)
