@@ -52,15 +52,31 @@ static float* f_ptr_mut(torch::Tensor& t) { return const_cast<float*>(f_ptr(t));
 // ---------------------------------------------------------------------------
 // C[M,N] = A' @ B' + bias. trans_a: A is [K,M] (weight-grad path);
 // trans_b: B is [K,N]; otherwise B is the NT weight layout [N,K].
+//
+// k_pad (NT only): run the reduction over k_pad >= K columns so the
+// kernel has no K-tail phase (a tail tile measured +13.5 us; see
+// PERF.md). Contract: B is the pre-padded [N, k_pad] tensor with ZERO
+// in columns [K, k_pad) — A's over-read garbage multiplies those zeros
+// — and A's storage extends >= 128 finite (zeroed) bytes past its end,
+// because rows over-read into the next row and the last (and clamped)
+// rows over-read into that slack. HipModel's slack-provisioned buffers
+// guarantee both; callers must not pass k_pad for arbitrary tensors.
 static void gemm(const torch::Tensor& A, const torch::Tensor& B,
                  torch::Tensor& C, const c10::optional<torch::Tensor>& bias,
-                 bool trans_a, bool trans_b) {
+                 bool trans_a, bool trans_b, int k_pad = 0) {
   int M = C.size(0), N = C.size(1);
   int K = trans_a ? A.size(0) : A.size(1);
   TORCH_CHECK((trans_a ? A.size(1) : A.size(0)) == M, "gemm: A/C M mismatch");
   TORCH_CHECK((trans_b ? B.size(1) : B.size(0)) == N, "gemm: B/C N mismatch");
-  TORCH_CHECK((trans_b ? B.size(0) : B.size(1)) == K, "gemm: A/B K mismatch");
+  if (k_pad) {
+    TORCH_CHECK(!trans_a && !trans_b, "gemm: k_pad is NT-only");
+    TORCH_CHECK(k_pad >= K && k_pad % 64 == 0, "gemm: bad k_pad");
+    TORCH_CHECK(B.size(1) == k_pad, "gemm: B not padded to k_pad");
+  } else {
+    TORCH_CHECK((trans_b ? B.size(0) : B.size(1)) == K, "gemm: A/B K mismatch");
+  }
   int lda = A.size(1), ldb = B.size(1), ldc = C.size(1);
+  if (k_pad) K = k_pad;  // lda stays the REAL row stride of A
   const float* bp = bias ? f_ptr(*bias) : nullptr;
   auto stream = current_stream();
   TORCH_CHECK(trans_a == trans_b, "gemm: only NT and TN layouts are wired");
@@ -428,7 +444,9 @@ static void transpose_bf16(const torch::Tensor& src, torch::Tensor& dst) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "zaremba_amd gfx950 kernel library";
-  m.def("gemm", &zamd::gemm, "MFMA bf16 GEMM (NT / TN)");
+  m.def("gemm", &zamd::gemm, "MFMA bf16 GEMM (NT / TN)", py::arg("A"),
+        py::arg("B"), py::arg("C"), py::arg("bias"), py::arg("trans_a"),
+        py::arg("trans_b"), py::arg("k_pad") = 0);
   m.def("lstm_seq_fwd", &zamd::lstm_seq_fwd);
   m.def("lstm_seq_bwd", &zamd::lstm_seq_bwd);
   m.def("lstm_cell_fwd_step", &zamd::lstm_cell_fwd_step);

@@ -28,16 +28,36 @@ def _ks(k):
     return (k + 31) // 32
 
 
+def _pad64(k):
+    return ((k + 63) // 64) * 64
+
+
+def _slacked(shape, dtype, device, slack_ptrs=None):
+    """Allocate a zeroed tensor with 64 extra zeroed elements of storage
+    slack after it. Tensors used as the A operand of a k_pad GEMM (see
+    ext_bind.hip:gemm) over-read up to 128 bytes past their last row;
+    the slack keeps that read in-bounds and finite (zero)."""
+    n = 1
+    for s in shape:
+        n *= s
+    flat = torch.zeros(n + 64, dtype=dtype, device=device)
+    if slack_ptrs is not None:
+        slack_ptrs.add(flat.untyped_storage().data_ptr())
+    return flat[:n].view(*shape)
+
+
 class _LayerWorkspace:
-    def __init__(self, T: int, B: int, H: int, device):
+    def __init__(self, T: int, B: int, H: int, device, slack_ptrs=None):
         bf, f32 = torch.bfloat16, torch.float32
         self.T, self.B, self.H = T, B, H
         self.gx = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
-        self.h_all = torch.zeros(T + 1, B, H, dtype=bf, device=device)
+        # h_all / dG are slack-provisioned: in eval mode h_all[1:] is the
+        # next consumer GEMM's A operand; dG is the dx GEMM's A operand.
+        self.h_all = _slacked((T + 1, B, H), bf, device, slack_ptrs)
         self.c_all = torch.zeros(T + 1, B, H, dtype=f32, device=device)
         self.gates = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
         self.dY = torch.zeros(T, B, H, dtype=bf, device=device)
-        self.dG = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
+        self.dG = _slacked((T, B, 4 * H), bf, device, slack_ptrs)
         self.dh_rec = torch.zeros(2, B, H, dtype=f32, device=device)
         self.dc = torch.zeros(B, H, dtype=f32, device=device)
         # fragment-packed workspaces (zero-prefilled: pad rows/K-tails
@@ -75,7 +95,14 @@ class _LayerRuntime:
         bf = torch.bfloat16
         self.Wx = torch.empty_like(layer.W_x, dtype=bf, device=device)
         self.Wh = torch.empty_like(layer.W_h, dtype=bf, device=device)
-        self.WxT = torch.empty(self.Hin, 4 * H, dtype=bf, device=device)
+        # K-padded dx-GEMM operand: WxT gets zero pad columns up to H4p
+        # for free from transpose_bf16's dst-stride, so the dx GEMM runs
+        # tail-free (measured ~2 us/launch on the 64-tile shapes; padding
+        # the forward GEMMs' Wx/fc.W was measured a net LOSS — the
+        # per-step padded-shadow copies cost ~15 us each, more than the
+        # 5.7 us tail they remove — and is not done).
+        self.H4p = _pad64(4 * H)
+        self.WxT = torch.zeros(self.Hin, self.H4p, dtype=bf, device=device)
         self.WhT = torch.empty(H, 4 * H, dtype=bf, device=device)
         # fragment-packed W_h (forward cell) and W_h^T (backward hop)
         nb = (H + 15) // 16
@@ -86,6 +113,7 @@ class _LayerRuntime:
         self.bias_sum = torch.empty(4 * H, dtype=torch.float32,
                                     device=device)
         self.ws: Optional[_LayerWorkspace] = None
+        self.slack_ptrs = None  # shared set, attached by HipModel
 
     @torch.no_grad()
     def refresh(self, layer, e):
@@ -97,9 +125,9 @@ class _LayerRuntime:
         e.pack_gated_w(self.WhT, self.WhTP, self.H, 1, 4 * self.H)
         torch.add(layer.b_x, layer.b_h, out=self.bias_sum)
 
-    def ensure_ws(self, T, B, device):
+    def ensure_ws(self, T, B, device, slack_ptrs=None):
         if self.ws is None or self.ws.T != T or self.ws.B != B:
-            self.ws = _LayerWorkspace(T, B, self.H, device)
+            self.ws = _LayerWorkspace(T, B, self.H, device, slack_ptrs)
         return self.ws
 
 
@@ -108,9 +136,14 @@ class _FcRuntime:
         bf = torch.bfloat16
         V, H = fc.W.shape
         self.W = torch.empty(V, H, dtype=bf, device=device)
-        self.WT = torch.empty(H, V, dtype=bf, device=device)
+        # K-padded backward-dx operand (see _LayerRuntime): WT carries
+        # zero pad columns up to Vp from its strided transpose refresh.
+        self.Vp = _pad64(V)
+        self.WT = torch.zeros(H, self.Vp, dtype=bf, device=device)
         self.dscT = None  # [V, TBp] zero-padded, sized on first backward
         self.xT = None    # [H, TBp]
+        self.dsc_buf = None  # slacked [N, V] bf16, sized on first backward
+        self.slack_ptrs = None  # shared set, attached by HipModel
 
     @torch.no_grad()
     def refresh(self, fc, e):
@@ -134,8 +167,15 @@ class HipModel:
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,
                                       device=dev)
+        # Storages with >=128 B of zeroed slack past the tensor end: only
+        # these may be the A operand of a k_pad GEMM (ext_bind.hip:gemm).
+        self.slack_ptrs = set()
+        self._slack_bufs = {}
         self.layers = [_LayerRuntime(l, dev) for l in model.rnns]
+        for rt in self.layers:
+            rt.slack_ptrs = self.slack_ptrs
         self.fc = _FcRuntime(model.fc, dev)
+        self.fc.slack_ptrs = self.slack_ptrs
         self.dropout_seed = secrets.randbits(63)
         self.dropout_counter = torch.zeros(1, dtype=torch.int64, device=dev)
         self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
@@ -161,6 +201,17 @@ class HipModel:
     def invalidate_shadows(self):
         self._shadows_fresh = False
 
+    def slack_buf(self, key, shape, dtype):
+        """Per-call-site cached activation buffer with zeroed storage
+        slack (valid to reuse every step: the training loop is strictly
+        fwd -> bwd -> step, the module-docstring aliasing contract)."""
+        k = (key, tuple(shape), dtype)
+        buf = self._slack_bufs.get(k)
+        if buf is None:
+            buf = _slacked(shape, dtype, self.device, self.slack_ptrs)
+            self._slack_bufs[k] = buf
+        return buf
+
     # ------------------------------------------------------------------
     def forward(self, x, states, training: bool):
         if not self._shadows_fresh:
@@ -169,14 +220,18 @@ class HipModel:
         T, B = x.shape
         p = m.dropout_p
         idx = x.reshape(-1)
-        emb = EmbeddingFn.apply(m.embed.W, idx, self.emb_W)
-        cur = emb.view(T, B, m.hidden_size)
+        H = m.hidden_size
+        bf = torch.bfloat16
+        emb = EmbeddingFn.apply(m.embed.W, idx, self.emb_W,
+                                [self.slack_buf("emb", (T * B, H), bf)])
+        cur = emb.view(T, B, H)
         if training and p > 0:
             cur = DropoutFn.apply(cur, p, self.dropout_seed,
-                                  self.dropout_counter)
+                                  self.dropout_counter,
+                                  [self.slack_buf(("drop", 0), cur.shape, bf)])
         new_states = list(states)
         for i, (rt, layer) in enumerate(zip(self.layers, m.rnns)):
-            rt.ensure_ws(T, B, self.device)
+            rt.ensure_ws(T, B, self.device, self.slack_ptrs)
             h0, c0 = states[i]
             out, hT, cT = LstmLayerFn.apply(cur, h0, c0, layer.W_x,
                                             layer.W_h, layer.b_x, layer.b_h,
@@ -185,7 +240,9 @@ class HipModel:
             cur = out
             if training and p > 0:
                 cur = DropoutFn.apply(cur, p, self.dropout_seed,
-                                      self.dropout_counter)
+                                      self.dropout_counter,
+                                      [self.slack_buf(("drop", i + 1),
+                                                      cur.shape, bf)])
         scores = LinearFn.apply(cur.reshape(T * B, m.hidden_size), m.fc.W,
                                 m.fc.b, self.fc)
         for i in range(len(new_states)):

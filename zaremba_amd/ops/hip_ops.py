@@ -27,10 +27,12 @@ class EmbeddingFn(torch.autograd.Function):
     """K1: gather fwd / fp32 scatter-add bwd (reference model.py:6-17)."""
 
     @staticmethod
-    def forward(ctx, W_master, idx, shadow_W):
+    def forward(ctx, W_master, idx, shadow_W, out=None):
+        # out: None or a 1-list [buf] (see DropoutFn)
         N = idx.numel()
         H = W_master.size(1)
-        out = torch.empty(N, H, dtype=torch.bfloat16, device=idx.device)
+        out = (torch.empty(N, H, dtype=torch.bfloat16, device=idx.device)
+               if out is None else out[0])
         ext().embedding_fwd(shadow_W, idx, out)
         ctx.save_for_backward(idx)
         ctx.V = W_master.size(0)
@@ -43,7 +45,7 @@ class EmbeddingFn(torch.autograd.Function):
         dW = torch.zeros(ctx.V, dY.size(1), dtype=torch.float32,
                          device=dY.device)
         ext().embedding_bwd(dY, idx, dW)
-        return dW, None, None
+        return dW, None, None, None
 
 
 class DropoutFn(torch.autograd.Function):
@@ -51,9 +53,11 @@ class DropoutFn(torch.autograd.Function):
     from (seed, saved_offset) — no mask tensor stored."""
 
     @staticmethod
-    def forward(ctx, x, p, seed, counter):
+    def forward(ctx, x, p, seed, counter, out=None):
+        # out: None or a 1-list [buf] (a non-Tensor holder, so autograd
+        # does not treat the reused buffer as an aliased input)
         x = x.contiguous()
-        y = torch.empty_like(x)
+        y = torch.empty_like(x) if out is None else out[0].view_as(x)
         saved_offset = torch.zeros(1, dtype=torch.int64, device=x.device)
         ext().dropout_fwd(x, y, p, seed, counter, saved_offset)
         ctx.p = p
@@ -67,7 +71,7 @@ class DropoutFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = torch.empty_like(dy)
         ext().dropout_bwd(dy, dx, ctx.p, ctx.seed, saved_offset)
-        return dx, None, None, None
+        return dx, None, None, None, None
 
 
 class LinearFn(torch.autograd.Function):
@@ -76,7 +80,7 @@ class LinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, W_master, b_master, fc_rt):
-        # x: [N, H] bf16; fc_rt holds shadows W (bf16 [V,H]) and WT ([H,V])
+        # x: [N, H] bf16; fc_rt holds shadows W (bf16 [V,H]) and WT
         x = x.contiguous()
         N, H = x.shape
         V = W_master.size(0)
@@ -91,11 +95,17 @@ class LinearFn(torch.autograd.Function):
         (x,) = ctx.saved_tensors
         fc_rt = ctx.fc_rt
         e = ext()
-        dsc = dscores.to(torch.bfloat16).contiguous()
         N, H = x.shape
-        V = dsc.size(1)
+        V = dscores.size(1)
+        if fc_rt.dsc_buf is None or fc_rt.dsc_buf.shape != (N, V):
+            from .hip_model import _slacked
+            fc_rt.dsc_buf = _slacked((N, V), torch.bfloat16, x.device,
+                                     fc_rt.slack_ptrs)
+        dsc = fc_rt.dsc_buf
+        dsc.copy_(dscores)
         dx = torch.empty(N, H, dtype=torch.bfloat16, device=x.device)
-        e.gemm(dsc, fc_rt.WT, dx, None, False, False)
+        e.gemm(dsc, fc_rt.WT, dx, None, False, False,
+               fc_rt.Vp if fc_rt.Vp != V else 0)
         # dW = dsc^T @ x via explicit transposes + the fast NT kernel (the
         # TN staging path is register-starved; measured 2-3x slower).
         # Transposed temps are zero-padded in K so the GEMM has no K-tail
@@ -207,7 +217,10 @@ class LstmLayerFn(torch.autograd.Function):
         dWx = torch.empty(4 * H, Hin, dtype=torch.float32, device=x2.device)
         e.gemm(dGT, xT, dWx, None, False, False)
         dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
-        e.gemm(dG2, rt.WxT, dx, None, False, False)
+        # dG is always slack-provisioned workspace; WxT carries zero pad
+        # columns up to H4p from its strided transpose refresh
+        e.gemm(dG2, rt.WxT, dx, None, False, False,
+               rt.H4p if rt.H4p != 4 * H else 0)
         db = torch.zeros(4 * H, dtype=torch.float32, device=x2.device)
         e.colsum_bf16(dG2, db)  # grads of b_x and b_h are identical
         return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db.clone(),
