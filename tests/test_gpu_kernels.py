@@ -73,6 +73,32 @@ def test_gemm_tn(ext, M, N, K):
     assert rel_err(C, ref) < 3e-2, rel_err(C, ref)
 
 
+@pytest.mark.parametrize("M,N,K,k_pad", [
+    (700, 1500, 6000, 6016),   # lstm dx shape (k_pad path)
+    (700, 1500, 10048, 0),     # proj dx shape, K already a BK multiple
+    (130, 70, 256, 0),         # M/N edge blocks
+])
+def test_gemm_splitk(ext, M, N, K, k_pad):
+    """2-way split-K partials + combine == plain NT GEMM result."""
+    torch.manual_seed(7)
+    A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
+    Keff = k_pad if k_pad else K
+    B = torch.zeros(N, Keff, device=dev(), dtype=torch.bfloat16)
+    B[:, :K].normal_()
+    if k_pad:
+        # k_pad contract: A needs >=128 B of finite storage slack
+        flat = torch.zeros(M * K + 64, device=dev(), dtype=torch.bfloat16)
+        flat[:M * K].copy_(A.reshape(-1))
+        A = flat[:M * K].view(M, K)
+    C1 = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    C2 = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    out = torch.empty(M, N, device=dev(), dtype=torch.bfloat16)
+    ext.gemm_splitk(A, B, C1, C2, None, k_pad)
+    ext.add2_f32_bf16(C1, C2, out)
+    ref = A.float() @ B[:, :K].float().t()
+    assert rel_err(out, ref) < 5e-2, rel_err(out, ref)
+
+
 def test_gemm_bf16_out(ext):
     torch.manual_seed(2)
     M, N, K = 130, 140, 96

@@ -33,3 +33,17 @@ bench(768, 6016, 1536, torch.bfloat16)
 # proj
 bench(700, 10000, 1500)
 bench(768, 10048, 1536)
+
+# --- split-K hypothesis probe: dx GEMMs run ~1 block/CU (264 WGs of
+# 64-tile); if doubling the grid doesn't double time, a 2-way K-split
+# (2 co-resident blocks/CU interleaving their latency chains) pays.
+if len(sys.argv) > 1 and sys.argv[1] == "splitk":
+    print("== grid-doubling probe (64-tile dx shapes) ==")
+    bench(700, 1500, 6016, torch.bfloat16)    # 264 WGs
+    bench(1400, 1500, 6016, torch.bfloat16)   # 528 WGs, 2x work
+    bench(2800, 1500, 6016, torch.bfloat16)   # 1056 WGs, 4x work
+    bench(700, 3000, 6016, torch.bfloat16)    # 528 WGs via N
+    bench(700, 1500, 3008, torch.bfloat16)    # half-K: lower bound/2way
+    bench(700, 1500, 10048, torch.bfloat16)   # proj dx
+    bench(700, 1500, 5024, torch.bfloat16)    # its half-K
+    bench(1400, 1500, 10048, torch.bfloat16)

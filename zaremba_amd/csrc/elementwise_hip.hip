@@ -463,4 +463,30 @@ void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                      R, C);
 }
 
+// combine the two f32 partials of a split-K GEMM into the bf16 result
+// (one rounding: partials carry the exact f32 MFMA accumulators).
+__global__ void add2_f32_bf16_kernel(const float* __restrict__ a,
+                                     const float* __restrict__ b,
+                                     bf16* __restrict__ out, int64_t n) {
+  int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (i + 8 <= n) {
+    float4 a0 = *(const float4*)(a + i), a1 = *(const float4*)(a + i + 4);
+    float4 b0 = *(const float4*)(b + i), b1 = *(const float4*)(b + i + 4);
+    bf16x8 v = {(bf16)(a0.x + b0.x), (bf16)(a0.y + b0.y),
+                (bf16)(a0.z + b0.z), (bf16)(a0.w + b0.w),
+                (bf16)(a1.x + b1.x), (bf16)(a1.y + b1.y),
+                (bf16)(a1.z + b1.z), (bf16)(a1.w + b1.w)};
+    *(bf16x8*)(out + i) = v;
+  } else {
+    for (; i < n; ++i) out[i] = (bf16)(a[i] + b[i]);
+  }
+}
+
+void launch_add2_f32_bf16(const float* a, const float* b, bf16* out,
+                          int64_t n, hipStream_t stream) {
+  int64_t thr = cdiv(n, 8);
+  hipLaunchKernelGGL(add2_f32_bf16_kernel, dim3(cdiv(thr, 256)), dim3(256),
+                     0, stream, a, b, out, n);
+}
+
 }  // namespace zamd

@@ -98,6 +98,38 @@ static void gemm(const torch::Tensor& A, const torch::Tensor& B,
   }
 }
 
+// 2-way split-K NT GEMM: f32 partials into C and C2 (the exact MFMA
+// accumulators of the two K halves), combined by add2_f32_bf16. Same
+// k_pad contract as gemm(); the split keys off the padded K so both
+// halves are whole-BK-tile ranges.
+static void gemm_splitk(const torch::Tensor& A, const torch::Tensor& B,
+                        torch::Tensor& C, torch::Tensor& C2,
+                        const c10::optional<torch::Tensor>& bias, int k_pad) {
+  int M = C.size(0), N = C.size(1);
+  int K = A.size(1);
+  TORCH_CHECK(A.size(0) == M, "gemm_splitk: A/C M mismatch");
+  TORCH_CHECK(B.size(0) == N, "gemm_splitk: B/C N mismatch");
+  int Keff = k_pad ? k_pad : K;
+  TORCH_CHECK(B.size(1) == Keff, "gemm_splitk: B/K mismatch");
+  TORCH_CHECK(Keff % 64 == 0 && Keff >= 128, "gemm_splitk: bad K");
+  TORCH_CHECK(C2.sizes() == C.sizes(), "gemm_splitk: C/C2 mismatch");
+  TORCH_CHECK(C.scalar_type() == torch::kFloat32 &&
+              C2.scalar_type() == torch::kFloat32,
+              "gemm_splitk: f32 partials expected");
+  const float* bp = bias ? f_ptr(*bias) : nullptr;
+  launch_gemm_splitk_t<float>(bf_ptr(A), bf_ptr(B), C.data_ptr<float>(),
+                              C2.data_ptr<float>(), bp, M, N, Keff,
+                              A.size(1), B.size(1), C.size(1),
+                              current_stream());
+}
+
+static void add2_f32_bf16(const torch::Tensor& a, const torch::Tensor& b,
+                          torch::Tensor& out) {
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() == out.numel());
+  launch_add2_f32_bf16(f_ptr(a), f_ptr(b), bf_ptr_mut(out), a.numel(),
+                       current_stream());
+}
+
 // ---------------------------------------------------------------------------
 // LSTM sequence drivers (+ hipGraph cache)
 // ---------------------------------------------------------------------------
@@ -447,6 +479,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &zamd::gemm, "MFMA bf16 GEMM (NT / TN)", py::arg("A"),
         py::arg("B"), py::arg("C"), py::arg("bias"), py::arg("trans_a"),
         py::arg("trans_b"), py::arg("k_pad") = 0);
+  m.def("gemm_splitk", &zamd::gemm_splitk,
+        "2-way split-K NT GEMM (f32 partials)", py::arg("A"), py::arg("B"),
+        py::arg("C"), py::arg("C2"), py::arg("bias"), py::arg("k_pad") = 0);
+  m.def("add2_f32_bf16", &zamd::add2_f32_bf16);
   m.def("lstm_seq_fwd", &zamd::lstm_seq_fwd);
   m.def("lstm_seq_bwd", &zamd::lstm_seq_bwd);
   m.def("lstm_cell_fwd_step", &zamd::lstm_cell_fwd_step);

@@ -173,7 +173,24 @@ template <int TILE, bool TRANS_A, bool TRANS_B, typename OutT>
 __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     OutT* __restrict__ C, const float* __restrict__ bias, int M, int N,
-    int K, int lda, int ldb, int ldc) {
+    int K, int lda, int ldb, int ldc, OutT* __restrict__ C2, int kt_split) {
+  // 2-way split-K (kt_split > 0): blockIdx.z picks a K-tile range and a
+  // partial output (z0 -> C with bias, z1 -> C2 without). 2 co-resident
+  // blocks/CU interleave their latency chains on the ~1-block/CU dx
+  // shapes (measured: doubling the grid is near-free, so the split runs
+  // ~2x faster than one full-K pass). Caller combines C + C2.
+  if (kt_split > 0) {
+    if (blockIdx.z) {
+      const int64_t koff = (int64_t)kt_split * BK;
+      A += koff;
+      B += koff;
+      K -= kt_split * BK;
+      C = C2;
+      bias = nullptr;
+    } else {
+      K = kt_split * BK;
+    }
+  }
   constexpr int WT = TILE / 2;       // wave tile (64 or 32)
   constexpr int NF = WT / 16;        // fragments per wave dim (4 or 2)
   // one LDS object only (a second __shared__ forces vmcnt(0) before every
@@ -323,13 +340,28 @@ void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
   if (grid128 >= 192) {
     hipLaunchKernelGGL((gemm_kernel<128, TA, TB, OutT>), dim3(grid128),
                        dim3(GEMM_THREADS), 0, stream, A, B, C, bias, M, N, K,
-                       lda, ldb, ldc);
+                       lda, ldb, ldc, (OutT*)nullptr, 0);
   } else {
     int grid64 = cdiv(M, 64) * cdiv(N, 64);
     hipLaunchKernelGGL((gemm_kernel<64, TA, TB, OutT>), dim3(grid64),
                        dim3(GEMM_THREADS), 0, stream, A, B, C, bias, M, N, K,
-                       lda, ldb, ldc);
+                       lda, ldb, ldc, (OutT*)nullptr, 0);
   }
+}
+
+// 2-way split-K NT GEMM writing partials C/C2 (caller combines). K must
+// be a BK multiple with >= 2 tiles (the k_pad contract guarantees it on
+// the dx shapes). 64-tile: the target shapes run ~1 block/CU, the z=2
+// grid puts 2 blocks/CU.
+template <typename OutT>
+void launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
+                          const float* bias, int M, int N, int K, int lda,
+                          int ldb, int ldc, hipStream_t stream) {
+  int grid64 = cdiv(M, 64) * cdiv(N, 64);
+  int kt_split = (K / BK) / 2;
+  hipLaunchKernelGGL((gemm_kernel<64, false, false, OutT>),
+                     dim3(grid64, 1, 2), dim3(GEMM_THREADS), 0, stream, A, B,
+                     C, bias, M, N, K, lda, ldb, ldc, C2, kt_split);
 }
 
 #define INST(TA, TB, T)                                                     \
@@ -341,5 +373,11 @@ INST(false, false, bf16)
 INST(true, true, float)
 INST(true, true, bf16)
 #undef INST
+template void launch_gemm_splitk_t<float>(const bf16*, const bf16*, float*,
+                                          float*, const float*, int, int,
+                                          int, int, int, int, hipStream_t);
+template void launch_gemm_splitk_t<bf16>(const bf16*, const bf16*, bf16*,
+                                         bf16*, const float*, int, int, int,
+                                         int, int, int, hipStream_t);
 
 }  // namespace zamd

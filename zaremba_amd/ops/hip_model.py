@@ -114,6 +114,7 @@ class _LayerRuntime:
                                     device=device)
         self.ws: Optional[_LayerWorkspace] = None
         self.slack_ptrs = None  # shared set, attached by HipModel
+        self.dx_part = None  # f32 split-K partial pair, sized on first bwd
 
     @torch.no_grad()
     def refresh(self, layer, e):
@@ -143,6 +144,7 @@ class _FcRuntime:
         self.dscT = None  # [V, TBp] zero-padded, sized on first backward
         self.xT = None    # [H, TBp]
         self.dsc_buf = None  # slacked [N, V] bf16, sized on first backward
+        self.dx_part = None  # f32 split-K partial pair, sized on first bwd
         self.slack_ptrs = None  # shared set, attached by HipModel
 
     @torch.no_grad()
