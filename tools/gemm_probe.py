@@ -47,3 +47,27 @@ if len(sys.argv) > 1 and sys.argv[1] == "splitk":
     bench(700, 1500, 10048, torch.bfloat16)   # proj dx
     bench(700, 1500, 5024, torch.bfloat16)    # its half-K
     bench(1400, 1500, 10048, torch.bfloat16)
+
+if len(sys.argv) > 1 and sys.argv[1] == "splitk2":
+    def bench_splitk(M, N, K, k_pad=0, iters=30):
+        Keff = k_pad if k_pad else K
+        A = torch.randn(M, Keff, device=dev, dtype=torch.bfloat16)[:, :K].contiguous()
+        flat = torch.zeros(M*K + 64, device=dev, dtype=torch.bfloat16)
+        flat[:M*K].copy_(A.reshape(-1)); A = flat[:M*K].view(M, K)
+        B = torch.zeros(N, Keff, device=dev, dtype=torch.bfloat16); B[:, :K].normal_()
+        C1 = torch.empty(M, N, device=dev, dtype=torch.float32)
+        C2 = torch.empty(M, N, device=dev, dtype=torch.float32)
+        out = torch.empty(M, N, device=dev, dtype=torch.bfloat16)
+        for _ in range(5):
+            ext.gemm_splitk(A, B, C1, C2, None, k_pad); ext.add2_f32_bf16(C1, C2, out)
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        for _ in range(iters):
+            ext.gemm_splitk(A, B, C1, C2, None, k_pad); ext.add2_f32_bf16(C1, C2, out)
+        torch.cuda.synchronize(); dt = (time.perf_counter()-t0)/iters
+        print(f"SK M={M:6d} N={N:6d} K={K:6d} (pad {Keff}): {dt*1e6:8.1f} us  {2*M*N*K/dt/1e12:7.1f} TF")
+    print("== split-K + combine vs plain ==")
+    bench_splitk(700, 1500, 6000, 6016)       # lstm dx (in-train now)
+    bench_splitk(700, 1500, 10000, 10048)     # proj dx
+    bench_splitk(700, 6000, 1500, 1536)       # input gemm candidate
+    bench(700, 6000, 1500, torch.bfloat16)    # plain comparison
+    bench(1400, 6000, 1536, torch.bfloat16)   # grid-doubling check

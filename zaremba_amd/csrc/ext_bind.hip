@@ -150,6 +150,14 @@ static bool g_use_graphs = true;
 static bool g_use_persistent = true;
 static bool g_use_persistent_bwd = false;  // see PERF.md: broadcast amplification loses
 
+static bool g_use_fused_bwd = true;
+
+// Fused backward step: every block pair must be co-resident (grid
+// 2*ceil(H/16) <= 256 CUs) and the batch fit the 32-row MFMA tile.
+static bool fused_bwd_ok(int B, int H) {
+  return g_use_fused_bwd && B <= 32 && ((H + 15) / 16) * 2 <= 256;
+}
+
 // The persistent forward needs every block co-resident and one cell
 // element per thread: B*HS <= 256, H even, B <= 32.
 static bool persistent_ok(int B, int H) {
@@ -257,6 +265,29 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
   const int HSp = persistent_hs(H);
   const int64_t rstep = (int64_t)((H + HSp - 1) / HSp) * B * 6 * HSp;
   float* dh2 = dh_rec + (int64_t)B * H;  // second K-slice partial
+  if (fused_bwd_ok(B, H) && T >= 2) {
+    // Fused path: dgate[T-1] standalone, then ONE launch per remaining
+    // step doing hop[t] + dgate[t-1] (see smallm_fused_bwd_kernel).
+    // dG_pack slots 0/1 alternate by step parity: a launch's phase-1
+    // readers must not see its phase-2 pack writes.
+    const int nbn = (H + 15) / 16;
+    unsigned int* flags = reinterpret_cast<unsigned int*>(hgran) + 32;
+    HIP_CHECK(hipMemsetAsync(flags, 0, nbn * sizeof(unsigned int), stream));
+    const int64_t pstride = (int64_t)((4 * H + 31) / 32) * 2 * 64 * 8;
+    launch_lstm_cell_bwd_elt(dY + (T - 1) * hstep, nullptr, dh2, dc,
+                             rec + (T - 1) * rstep, dG + (T - 1) * gstep,
+                             dG_pack, B, H, HSp, stream);
+    for (int t = T - 1; t >= 1; --t) {
+      const int i = T - 1 - t;
+      bf16* rd = dG_pack + (i & 1) * pstride;
+      bf16* wr = dG_pack + ((i + 1) & 1) * pstride;
+      launch_smallm_fused_bwd(rd, WT_pack, dh_rec, dY + (t - 1) * hstep, dc,
+                              rec + (t - 1) * rstep, dG + (t - 1) * gstep,
+                              wr, flags, abort_flag, B, H, 4 * H, HSp,
+                              (unsigned int)(T - t), stream);
+    }
+    return;  // hop[0]'s dh output is unused (truncated-BPTT detach)
+  }
   for (int t = T - 1; t >= 0; --t) {
     launch_lstm_cell_bwd_elt(dY + t * hstep,
                              (t == T - 1) ? nullptr : dh_rec, dh2, dc,
@@ -290,9 +321,20 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
   float* dcp = f_ptr_mut(dc);
   auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
   auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
+  if (fused_bwd_ok(B, H) && T >= 2) {
+    // pair counters live at hgran uint32[32 ..); 2 pack slots needed
+    TORCH_CHECK(hgran.numel() * 8 >= 128 + ((H + 15) / 16) * 4,
+                "lstm_seq_bwd: hgran too small for fused-bwd counters");
+    TORCH_CHECK(dG_pack.dim() == 2 && dG_pack.size(0) >= 2,
+                "lstm_seq_bwd: fused path needs >= 2 dG_pack slots");
+  }
   auto stream = current_stream();
-  if (!g_use_graphs || (g_use_persistent_bwd && persistent_ok(B, H) &&
-                        persistent_bwd_lds(B, H) <= 160 * 1024)) {
+  // Spin-synchronized kernels (fused pair / persistent) launch eagerly:
+  // hipGraph replay of cross-block-waiting kernels hangs intermittently
+  // on ROCm 7.x (PERF.md).
+  if (!g_use_graphs || (fused_bwd_ok(B, H) && T >= 2) ||
+      (g_use_persistent_bwd && persistent_ok(B, H) &&
+       persistent_bwd_lds(B, H) <= 160 * 1024)) {
     lstm_seq_bwd_body(dyp, gp, rp, cp, whtp, wtp, dgp, dgpk, dhp, dcp, hg,
                       ab, T, B, H, stream);
     return;
@@ -320,6 +362,7 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
 static void set_use_graphs(bool v) { g_use_graphs = v; }
 static void set_use_persistent(bool v) { g_use_persistent = v; }
 static void set_use_persistent_bwd(bool v) { g_use_persistent_bwd = v; }
+static void set_use_fused_bwd(bool v) { g_use_fused_bwd = v; }
 static void clear_graphs() {
   for (auto& kv : g_fwd_graphs.cache) hipGraphExecDestroy(kv.second);
   for (auto& kv : g_bwd_graphs.cache) hipGraphExecDestroy(kv.second);
@@ -504,5 +547,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_use_persistent", &zamd::set_use_persistent);
   m.def("persistent_hs", &zamd::persistent_hs);
   m.def("set_use_persistent_bwd", &zamd::set_use_persistent_bwd);
+  m.def("set_use_fused_bwd", &zamd::set_use_fused_bwd);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

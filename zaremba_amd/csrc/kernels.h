@@ -40,6 +40,12 @@ void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec,
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
                              float* C, float* C2, int M, int N, int K,
                              hipStream_t stream);
+void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
+                             float* P, const bf16* dy, float* dc,
+                             const bf16* rec, bf16* dG, bf16* dG_pack_out,
+                             unsigned int* flags, unsigned int* abort_flag,
+                             int M, int N, int K, int HSp, unsigned int step,
+                             hipStream_t stream);
 
 // lstm_persistent.hip — one launch for a whole layer unroll
 int persistent_hs(int H);

@@ -340,6 +340,196 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused backward step: recurrent hop for step t + dgate elementwise for
+// step t-1 in ONE launch (reference main.py:113 BPTT backward).
+// ---------------------------------------------------------------------------
+// The per-step pair (dgate kernel at the ~4.5 us dispatch floor + hop
+// kernel) costs ~13 us/step-layer; this kernel halves the launch count.
+// The fusion is column-aligned so NO freshly-produced data is broadcast
+// inside the launch (the persistent-backward failure mode, PERF.md):
+//   * phase 1 (hop): identical to smallm_packed_nt<32,2> — the block's
+//     K-half MFMA over the PREVIOUS launch's packed dG[t] (crosses a
+//     kernel boundary: bulk HBM-speed reads),
+//   * the pair of blocks sharing an n-tile exchange their two K-half
+//     partials through write-through f32 stores + one monotonic
+//     per-tile arrival counter (no grid barrier, no L2 invalidate:
+//     1.3 KB of fresh data read memory-side),
+//   * phase 2 (dgate[t-1]): each block of the pair takes half the batch
+//     rows of ITS OWN 16 columns — everything it needs (dh, rec, dY,
+//     dc) is block-local — and writes dG[t-1] + the packed slot the
+//     NEXT launch reads (double-buffered by step parity: this launch's
+//     phase-1 readers must not see this launch's phase-2 writes).
+typedef __attribute__((address_space(1))) unsigned int gau32;
+#define ZRLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+
+DEV_INLINE void store_wt_f32(float* p, float v) {
+  __hip_atomic_store((gau32*)(uintptr_t)p,
+                     __builtin_bit_cast(unsigned int, v), ZRLX_AGENT);
+}
+DEV_INLINE float load_wt_f32(const float* p) {
+  unsigned int u =
+      __hip_atomic_load((const gau32*)(uintptr_t)p, ZRLX_AGENT);
+  return __builtin_bit_cast(float, u);
+}
+
+template <int MAXB>
+__global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
+    const bf16* __restrict__ A_pack,  // packed dG[t] (previous launch)
+    const bf16* __restrict__ W_pack,  // packed W_h^T shadow
+    float* __restrict__ P,            // [2][M*N] f32 K-half partials
+    const bf16* __restrict__ dy,      // dY[t-1], [B,H]
+    float* __restrict__ dc,           // [B,H] carried cell grad
+    const bf16* __restrict__ rec,     // rec[t-1] block records
+    bf16* __restrict__ dG,            // dG[t-1] out, [B,4H]
+    bf16* __restrict__ dG_pack_out,   // packed slot for the NEXT launch
+    unsigned int* __restrict__ flags, // [ceil(N/16)] monotonic counters
+    unsigned int* __restrict__ abort_flag,
+    int M, int N, int K, int HSp, unsigned int step) {
+  __shared__ float red[4 * MAXB * 16];
+  __shared__ int ok_s;
+
+  const int nbn = (N + 15) / 16;
+  const int nb = blockIdx.x % nbn;
+  const int n0 = nb * 16;
+  const int sk = blockIdx.x / nbn;  // K half (0/1)
+  const int w = wave_id();
+  const int l = lane_id();
+  const int lm = l & 15;
+  const int KS = (K + 31) / 32;
+  const int KH = (KS + 1) / 2;
+  const int ks0 = sk * KH;
+  const int ks1 = min(ks0 + KH, KS);
+
+  const bf16x8* pa = reinterpret_cast<const bf16x8*>(A_pack) + l;
+  const bf16x8* pw =
+      reinterpret_cast<const bf16x8*>(W_pack) + (int64_t)nb * KS * 64 + l;
+
+  f32x4 acc0 = {}, acc1 = {};
+  const int nown = (ks1 - ks0 - w + 3) / 4;
+  int i = 0;
+  for (; i + 8 <= nown; i += 8) {
+    bf16x8 a0v[8], a1v[8], bwv[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int ks = ks0 + w + 4 * (i + u);
+      a0v[u] = pa[ks * 128];
+      a1v[u] = pa[ks * 128 + 64];
+      bwv[u] = pw[ks * 64];
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      acc0 = mfma_16x16x32_bf16(a0v[u], bwv[u], acc0);
+      acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
+    }
+  }
+  for (; i < nown; ++i) {
+    const int ks = ks0 + w + 4 * i;
+    bf16x8 a0v = pa[ks * 128];
+    bf16x8 a1v = pa[ks * 128 + 64];
+    bf16x8 bwv = pw[ks * 64];
+    acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
+    acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
+  }
+
+  const int fr0 = (l >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    red[(w * MAXB + fr0 + r) * 16 + lm] = acc0[r];
+    red[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+  }
+  __syncthreads();
+
+  // publish this K-half's partial (write-through: memory-side visible
+  // once the wave's vmcnt drains; the pair partner reads it sc1)
+  float* mine = P + (int64_t)sk * M * N;
+  for (int idx = threadIdx.x; idx < M * 16; idx += CELL_THREADS) {
+    const int b = idx / 16, jj = idx % 16;
+    if (n0 + jj >= N) continue;
+    float v = red[(0 * MAXB + b) * 16 + jj] + red[(1 * MAXB + b) * 16 + jj] +
+              red[(2 * MAXB + b) * 16 + jj] + red[(3 * MAXB + b) * 16 + jj];
+    store_wt_f32(mine + (int64_t)b * N + n0 + jj, v);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    ok_s = 1;
+    gau32* f = (gau32*)(uintptr_t)(flags + nb);
+    __hip_atomic_fetch_add(f, 1u, ZRLX_AGENT);
+    unsigned int spins = 0;
+    while (__hip_atomic_load(f, ZRLX_AGENT) < 2u * step) {
+      __builtin_amdgcn_s_sleep(2);
+      if (++spins > 20000000u) {
+        atomicOr(abort_flag, 1u);
+        ok_s = 0;
+        break;
+      }
+    }
+  }
+  __syncthreads();
+  if (!ok_s) return;
+
+  // phase 2: dgate[t-1] for this block's half of the batch rows of its
+  // own 16 columns (mirrors lstm_cell_bwd_elt_kernel's math)
+  const float* partner = P + (int64_t)(sk ^ 1) * M * N;
+  const int Bh = (M + 1) / 2;
+  const int rb0 = sk * Bh;
+  const int nrows = min(M - rb0, Bh);
+  for (int idx = threadIdx.x; idx < nrows * 16; idx += CELL_THREADS) {
+    const int b = rb0 + idx / 16;
+    const int jj = idx % 16;
+    const int j = n0 + jj;
+    if (j >= N) continue;
+    const float vo = red[(0 * MAXB + b) * 16 + jj] +
+                     red[(1 * MAXB + b) * 16 + jj] +
+                     red[(2 * MAXB + b) * 16 + jj] +
+                     red[(3 * MAXB + b) * 16 + jj];
+    const int64_t e = (int64_t)b * N + j;
+    const float dh_rec = vo + load_wt_f32(partner + e);
+    const int blk = j / HSp, jr = j % HSp;
+    const bf16* r = rec + (((int64_t)blk * M + b) * 6) * HSp;
+    const float i_ = bf2f(r[0 * HSp + jr]);
+    const float f_ = bf2f(r[1 * HSp + jr]);
+    const float o_ = bf2f(r[2 * HSp + jr]);
+    const float n_ = bf2f(r[3 * HSp + jr]);
+    const float tc = bf2f(r[4 * HSp + jr]);
+    const float cprev = bf2f(r[5 * HSp + jr]);
+    const float dh = bf2f(dy[e]) + dh_rec;
+    const float do_ = dh * tc;
+    const float dct = dc[e] + dh * o_ * (1.f - tc * tc);
+    const float di = dct * n_;
+    const float df = dct * cprev;
+    const float dn = dct * i_;
+    const bf16 v[4] = {f2bf(di * i_ * (1.f - i_)),
+                       f2bf(df * f_ * (1.f - f_)),
+                       f2bf(do_ * o_ * (1.f - o_)),
+                       f2bf(dn * (1.f - n_ * n_))};
+    const int64_t gbase = (int64_t)b * 4 * N + j;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int k = g * N + j;
+      dG[gbase + (int64_t)g * N] = v[g];
+      const int ks = k / 32, sub = k % 32;
+      const int pl = (b & 15) + 16 * (sub / 8);
+      dG_pack_out[(((int64_t)ks * 2 + b / 16) * 64 + pl) * 8 + sub % 8] =
+          v[g];
+    }
+    dc[e] = dct * f_;
+  }
+}
+
+void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
+                             float* P, const bf16* dy, float* dc,
+                             const bf16* rec, bf16* dG, bf16* dG_pack_out,
+                             unsigned int* flags, unsigned int* abort_flag,
+                             int M, int N, int K, int HSp, unsigned int step,
+                             hipStream_t stream) {
+  hipLaunchKernelGGL((smallm_fused_bwd_kernel<32>), dim3(cdiv(N, 16) * 2),
+                     dim3(CELL_THREADS), 0, stream, A_pack, W_pack, P, dy,
+                     dc, rec, dG, dG_pack_out, flags, abort_flag, M, N, K,
+                     HSp, step);
+}
+
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
                              float* C, float* C2, int M, int N, int K,
                              hipStream_t stream) {
