@@ -463,6 +463,26 @@ void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                      R, C);
 }
 
+// f32 column sum (projection db = colsum(dscores), f32 [N,V] input):
+// same row-chunked atomicAdd scheme as the bf16 variant.
+__global__ void colsum_f32_kernel(const float* __restrict__ in,
+                                  float* __restrict__ out, int Rr, int Cc) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= Cc) return;
+  int r0 = blockIdx.y * 64;
+  int r1 = min(r0 + 64, Rr);
+  float acc = 0.f;
+  for (int r = r0; r < r1; ++r) acc += in[(int64_t)r * Cc + c];
+  atomicAdd(out + c, acc);
+}
+
+void launch_colsum_f32(const float* in, float* out, int R, int C,
+                       hipStream_t stream) {
+  dim3 grid(cdiv(C, 256), cdiv(R, 64));
+  hipLaunchKernelGGL(colsum_f32_kernel, grid, dim3(256), 0, stream, in, out,
+                     R, C);
+}
+
 // combine the two f32 partials of a split-K GEMM into the bf16 result
 // (one rounding: partials carry the exact f32 MFMA accumulators).
 __global__ void add2_f32_bf16_kernel(const float* __restrict__ a,

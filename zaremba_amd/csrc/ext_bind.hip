@@ -505,6 +505,12 @@ static void colsum_bf16(const torch::Tensor& in, torch::Tensor& out) {
   launch_colsum_bf16(bf_ptr(in), f_ptr_mut(out), R, C, current_stream());
 }
 
+// db = colsum(dscores) for f32 [N, V] scores grads (out pre-zeroed)
+static void colsum_f32(const torch::Tensor& in, torch::Tensor& out) {
+  launch_colsum_f32(f_ptr(in), f_ptr_mut(out), in.size(0), in.size(1),
+                    current_stream());
+}
+
 // dst is [C, ldd] with ldd >= R; columns [R, ldd) are left untouched
 // (pre-zero them once when using the pad as a GEMM K extension).
 static void transpose_bf16(const torch::Tensor& src, torch::Tensor& dst) {
@@ -543,10 +549,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &zamd::sgd_update);
   m.def("transpose_bf16", &zamd::transpose_bf16);
   m.def("colsum_bf16", &zamd::colsum_bf16);
+  m.def("colsum_f32", &zamd::colsum_f32);
   m.def("set_use_graphs", &zamd::set_use_graphs);
   m.def("set_use_persistent", &zamd::set_use_persistent);
   m.def("persistent_hs", &zamd::persistent_hs);
   m.def("set_use_persistent_bwd", &zamd::set_use_persistent_bwd);
   m.def("set_use_fused_bwd", &zamd::set_use_fused_bwd);
+  m.def("fused_bwd_active", &zamd::fused_bwd_ok);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

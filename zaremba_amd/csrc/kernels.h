@@ -94,5 +94,7 @@ void launch_add2_f32_bf16(const float* a, const float* b, bf16* out,
                           int64_t n, hipStream_t stream);
 void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                         hipStream_t stream);
+void launch_colsum_f32(const float* in, float* out, int R, int C,
+                       hipStream_t stream);
 
 }  // namespace zamd

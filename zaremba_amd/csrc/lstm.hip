@@ -432,6 +432,32 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
   }
 
+  // Prefetch the dgate phase's partner-independent inputs (rec record,
+  // dY, dc) NOW: the loads complete under the publish/arrive/spin that
+  // follows instead of serializing into phase 2. nrows*16 <= 256, so
+  // each thread owns at most one dgate element.
+  const int Bh = (M + 1) / 2;
+  const int rb0 = sk * Bh;
+  const int nrows = min(M - rb0, Bh);
+  const int db = rb0 + threadIdx.x / 16;
+  const int dj = n0 + threadIdx.x % 16;
+  const bool dwork = ((int)threadIdx.x < nrows * 16) && (dj < N);
+  float p_i = 0.f, p_f = 0.f, p_o = 0.f, p_n = 0.f, p_tc = 0.f,
+        p_cprev = 0.f, p_dy = 0.f, p_dc = 0.f;
+  if (dwork) {
+    const int blk = dj / HSp, jr = dj % HSp;
+    const bf16* r = rec + (((int64_t)blk * M + db) * 6) * HSp;
+    p_i = bf2f(r[0 * HSp + jr]);
+    p_f = bf2f(r[1 * HSp + jr]);
+    p_o = bf2f(r[2 * HSp + jr]);
+    p_n = bf2f(r[3 * HSp + jr]);
+    p_tc = bf2f(r[4 * HSp + jr]);
+    p_cprev = bf2f(r[5 * HSp + jr]);
+    const int64_t e = (int64_t)db * N + dj;
+    p_dy = bf2f(dy[e]);
+    p_dc = dc[e];
+  }
+
   const int fr0 = (l >> 4) * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -470,40 +496,26 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   if (!ok_s) return;
 
   // phase 2: dgate[t-1] for this block's half of the batch rows of its
-  // own 16 columns (mirrors lstm_cell_bwd_elt_kernel's math)
-  const float* partner = P + (int64_t)(sk ^ 1) * M * N;
-  const int Bh = (M + 1) / 2;
-  const int rb0 = sk * Bh;
-  const int nrows = min(M - rb0, Bh);
-  for (int idx = threadIdx.x; idx < nrows * 16; idx += CELL_THREADS) {
-    const int b = rb0 + idx / 16;
-    const int jj = idx % 16;
-    const int j = n0 + jj;
-    if (j >= N) continue;
+  // own 16 columns (mirrors lstm_cell_bwd_elt_kernel's math; inputs
+  // preloaded above, only the partner partial is read here)
+  if (dwork) {
+    const float* partner = P + (int64_t)(sk ^ 1) * M * N;
+    const int b = db, j = dj, jj = dj - n0;
     const float vo = red[(0 * MAXB + b) * 16 + jj] +
                      red[(1 * MAXB + b) * 16 + jj] +
                      red[(2 * MAXB + b) * 16 + jj] +
                      red[(3 * MAXB + b) * 16 + jj];
     const int64_t e = (int64_t)b * N + j;
-    const float dh_rec = vo + load_wt_f32(partner + e);
-    const int blk = j / HSp, jr = j % HSp;
-    const bf16* r = rec + (((int64_t)blk * M + b) * 6) * HSp;
-    const float i_ = bf2f(r[0 * HSp + jr]);
-    const float f_ = bf2f(r[1 * HSp + jr]);
-    const float o_ = bf2f(r[2 * HSp + jr]);
-    const float n_ = bf2f(r[3 * HSp + jr]);
-    const float tc = bf2f(r[4 * HSp + jr]);
-    const float cprev = bf2f(r[5 * HSp + jr]);
-    const float dh = bf2f(dy[e]) + dh_rec;
-    const float do_ = dh * tc;
-    const float dct = dc[e] + dh * o_ * (1.f - tc * tc);
-    const float di = dct * n_;
-    const float df = dct * cprev;
-    const float dn = dct * i_;
-    const bf16 v[4] = {f2bf(di * i_ * (1.f - i_)),
-                       f2bf(df * f_ * (1.f - f_)),
-                       f2bf(do_ * o_ * (1.f - o_)),
-                       f2bf(dn * (1.f - n_ * n_))};
+    const float dh = p_dy + (vo + load_wt_f32(partner + e));
+    const float do_ = dh * p_tc;
+    const float dct = p_dc + dh * p_o * (1.f - p_tc * p_tc);
+    const float di = dct * p_n;
+    const float df = dct * p_cprev;
+    const float dn = dct * p_i;
+    const bf16 v[4] = {f2bf(di * p_i * (1.f - p_i)),
+                       f2bf(df * p_f * (1.f - p_f)),
+                       f2bf(do_ * p_o * (1.f - p_o)),
+                       f2bf(dn * (1.f - p_n * p_n))};
     const int64_t gbase = (int64_t)b * 4 * N + j;
 #pragma unroll
     for (int g = 0; g < 4; ++g) {
@@ -514,7 +526,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
       dG_pack_out[(((int64_t)ks * 2 + b / 16) * 64 + pl) * 8 + sub % 8] =
           v[g];
     }
-    dc[e] = dct * f_;
+    dc[e] = dct * p_f;
   }
 }
 
