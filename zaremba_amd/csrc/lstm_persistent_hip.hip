@@ -103,6 +103,25 @@ DEV_INLINE void store_pair_wt(bf16* p, bf16 lo, bf16 hi) {
   __hip_atomic_store((gu32*)(uintptr_t)p, packed, RLX_AGENT);
 }
 
+// Optional phase census (tools/fwd_census.hip compiles this file with
+// -DZAMD_FWD_PROF): s_memrealtime (100 MHz) deltas per phase per block,
+// accumulated into a device global. Compiled out of the production .so.
+#ifdef ZAMD_FWD_PROF
+__device__ unsigned long long g_fwd_prof[256 * 8];
+#define PROF_STAMP(v) \
+  unsigned long long v = \
+      (threadIdx.x == 0) ? __builtin_amdgcn_s_memrealtime() : 0
+#define PROF_ACC(ph, t0, t1) \
+  if (threadIdx.x == 0) g_fwd_prof[blockIdx.x * 8 + (ph)] += (t1) - (t0)
+#else
+#define PROF_STAMP(v) \
+  do {                \
+  } while (0)
+#define PROF_ACC(ph, t0, t1) \
+  do {                       \
+  } while (0)
+#endif
+
 // ===========================================================================
 // Forward
 // ===========================================================================
@@ -170,6 +189,9 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
   const int lk = (l >> 4) * 8;
 
   for (int t = 0; t < T; ++t) {
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt0);
+#endif
     // Prefetch this thread's gx slice BEFORE the barrier: the loads are
     // independent of h_t, and gx[t] is LLC-cold scattered data whose
     // latency otherwise lands inside the pointwise phase.
@@ -183,31 +205,56 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       gxo = gx[gxb + 2 * H];
       gxn = gx[gxb + 3 * H];
     }
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt1);
+    PROF_ACC(0, pt0, pt1);  // gx prefetch issue
+#endif
     if (t > 0) {
       if (!xcd_grid_barrier(pstate, NB, (unsigned int)t, abort_flag))
         return;
     } else {
       __syncthreads();
     }
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt2);
+    PROF_ACC(1, pt1, pt2);  // barrier (arrival drain + spin + acquire)
+#endif
     // ---- stage h_t into the LDS image (plain loads, L2-amplified) --------
+    // 8-vector register chunks: all 8 loads issue before any ds_write
+    // consumes one. The previous load-then-write-per-vector loop paid
+    // one vmcnt wait per 16 B — 7.3 us/step-layer, HALF the kernel
+    // (tools/fwd_census). Loads are unguarded vec8: the last vector of
+    // a row over-reads <= 14 B into the next h_all row (in-bounds: the
+    // staging reads slots 0..T-1 of the T+1-slot buffer), and the hs
+    // columns >= H it fills multiply Ws pad zeros in the MFMA.
     {
       const bf16* hsrc = h_all + (int64_t)t * B * H;
       const int vecs = (H + 7) / 8;
-      for (int idx = t_; idx < B * vecs; idx += PCELL_THREADS) {
-        const int b = idx / vecs;
-        const int k = (idx % vecs) * 8;
-        bf16x8 v = {};
-        const bf16* p = hsrc + (int64_t)b * H + k;
-        if (k + 8 <= H) {
-          v = *reinterpret_cast<const bf16x8*>(p);
-        } else {
+      const int total = B * vecs;
+      for (int idx = t_; idx < total; idx += 8 * PCELL_THREADS) {
+        bf16x8 v[8];
+        int id2 = idx;
 #pragma unroll
-          for (int e = 0; e < 8; ++e) v[e] = (k + e < H) ? p[e] : (bf16)0.f;
+        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+          const int ic = id2 < total ? id2 : total - 1;
+          const int b = ic / vecs, k = (ic % vecs) * 8;
+          v[u] = *reinterpret_cast<const bf16x8*>(hsrc + (int64_t)b * H + k);
         }
-        *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = v;
+        id2 = idx;
+#pragma unroll
+        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+          if (id2 < total) {
+            const int b = id2 / vecs, k = (id2 % vecs) * 8;
+            *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = v[u];
+          }
+        }
       }
     }
     __syncthreads();
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt3);
+    PROF_ACC(2, pt2, pt3);  // stage h into LDS
+#endif
 
     // ---- gate MFMA reduction (wave g -> gate g) --------------------------
     f32x4 acc0 = {}, acc1 = {};
@@ -246,6 +293,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       if (16 + fr0 + r < B) gbuf[(g * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
     }
     __syncthreads();
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt4);
+    PROF_ACC(3, pt3, pt4);  // gate MFMA + gbuf exchange
+#endif
 
     // ---- pointwise cell update ------------------------------------------
     if (own) {
@@ -276,6 +327,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       if (t == T - 1) c_all[((int64_t)T * B + b) * H + j] = c_reg;
     }
     __syncthreads();
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt5);
+    PROF_ACC(4, pt4, pt5);  // pointwise + rec stores
+#endif
     // ---- publish h_{t+1}: paired write-through stores --------------------
     {
       bf16* hdst = h_all + (int64_t)(t + 1) * B * H;
@@ -289,6 +344,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
         }
       }
     }
+#ifdef ZAMD_FWD_PROF
+    PROF_STAMP(pt6);
+    PROF_ACC(5, pt5, pt6);  // h publish issue
+#endif
   }
 }
 
