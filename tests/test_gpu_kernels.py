@@ -174,7 +174,7 @@ def _run_seq_fwd(ext, T, B, H, seed=5):
     gates = torch.zeros(T, B, 4 * H, device=dev(), dtype=torch.bfloat16)
     rec = torch.zeros(T * nb * B * 6 * hs, device=dev(),
                       dtype=torch.bfloat16)
-    hgran = torch.zeros(32, device=dev(), dtype=torch.int64)
+    hgran = torch.zeros(768, device=dev(), dtype=torch.int64)
     abort = torch.zeros(1, device=dev(), dtype=torch.int32)
     h_all[0] = (torch.randn(B, H, device=dev()) * 0.3).to(torch.bfloat16)
     c_all[0] = torch.randn(B, H, device=dev()) * 0.3

@@ -31,7 +31,7 @@ __global__ void census_kernel(unsigned int* cnt, unsigned int* result,
 }
 
 // the same two-level barrier as lstm_persistent.hip, iterated
-template <int SLEEP, int VARIANT>  // VARIANT 0: two-level; 1: flat top-counter poll; 2: two-level no-acquire (timing only)
+template <int SLEEP, int VARIANT>  // VARIANT 0: two-level; 1: flat top-counter poll; 2: two-level no-acquire (timing only); 3: two-level PADDED (each group counter/gen on its own 128B line) no-acquire; 4: flat no-acquire
 __device__ bool xcd_barrier(unsigned int* pstate, int grp, int nbg,
                             int ngroups, int nb, unsigned int gen,
                             unsigned int* fail) {
@@ -41,14 +41,32 @@ __device__ bool xcd_barrier(unsigned int* pstate, int grp, int nbg,
   if (threadIdx.x == 0) {
     ok_s = 1;
     gu32* st = (gu32*)(uintptr_t)pstate;
-    if (VARIANT == 1) {
+    if (VARIANT == 1 || VARIANT == 4) {
       __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
       unsigned int spins = 0;
       while (__hip_atomic_load(&st[8], RLX_AGENT) < gen * nb) {
         __builtin_amdgcn_s_sleep(SLEEP);
         if (++spins > 3000000u) { atomicAdd(fail, 1u); ok_s = 0; break; }
       }
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      if (VARIANT == 1)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    } else if (VARIANT == 3) {
+      // padded two-level: group counter at st[grp*32], top at st[256],
+      // per-group generation at st[288 + grp*32] — every hot word on its
+      // own 128-B line, so the 8 group add-chains run in parallel
+      unsigned int t = __hip_atomic_fetch_add(&st[512 + grp * 32], 1u, RLX_AGENT);
+      if (t == gen * nbg - 1) {
+        unsigned int tt = __hip_atomic_fetch_add(&st[768 + 256], 1u, RLX_AGENT);
+        if (tt == gen * ngroups - 1) {
+          for (int x = 0; x < 8; ++x)
+            __hip_atomic_store(&st[1056 + x * 32], gen, RLX_AGENT);
+        }
+      }
+      unsigned int spins = 0;
+      while (__hip_atomic_load(&st[1056 + grp * 32], RLX_AGENT) < gen) {
+        __builtin_amdgcn_s_sleep(SLEEP);
+        if (++spins > 3000000u) { atomicAdd(fail, 1u); ok_s = 0; break; }
+      }
     } else {
       unsigned int t = __hip_atomic_fetch_add(&st[grp], 1u, RLX_AGENT);
       if (t == gen * nbg - 1) {
@@ -91,9 +109,9 @@ int main() {
   size_t lds = (size_t)(4 * 6 + 20) * 1512 * 2 + 4 * 32 * 16 * 4;
   printf("LDS request: %zu bytes\n", lds);
   unsigned int* buf;
-  (void)hipMalloc(&buf, 4096);
+  (void)hipMalloc(&buf, 16384);
   for (int nb : {128, 200, 240, 248, 250, 252, 256, 260}) {
-    (void)hipMemset(buf, 0, 4096);
+    (void)hipMemset(buf, 0, 16384);
     hipLaunchKernelGGL(census_kernel, dim3(nb), dim3(256), lds, 0, buf,
                        buf + 64, nb);
     (void)hipDeviceSynchronize();
@@ -104,7 +122,7 @@ int main() {
   // barrier probe at the persistent-kernel sizes (timed)
   for (int nb : {188, 125, 94}) {
     auto run = [&](const char* name, auto kern) {
-      (void)hipMemset(buf, 0, 4096);
+      (void)hipMemset(buf, 0, 16384);
       hipEvent_t e0, e1;
       (void)hipEventCreate(&e0);
       (void)hipEventCreate(&e1);
@@ -125,6 +143,11 @@ int main() {
     run("2lvl sleep32", barrier_probe_kernel<32, 0>);
     run("flat sleep8", barrier_probe_kernel<8, 1>);
     run("2lvl sleep8 noacq", barrier_probe_kernel<8, 2>);
+    run("flat sleep8 noacq", barrier_probe_kernel<8, 4>);
+    run("flat sleep2 noacq", barrier_probe_kernel<2, 4>);
+    run("PADDED2lvl sl8 noacq", barrier_probe_kernel<8, 3>);
+    run("PADDED2lvl sl2 noacq", barrier_probe_kernel<2, 3>);
+    run("PADDED2lvl sl1 noacq", barrier_probe_kernel<1, 3>);
   }
   return 0;
 }

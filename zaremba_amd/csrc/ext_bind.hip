@@ -173,7 +173,7 @@ static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,
                               unsigned int* abort_flag, int T, int B, int H,
                               hipStream_t stream) {
   if (persistent_ok(B, H)) {
-    HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
+    HIP_CHECK(hipMemsetAsync(hgran, 0, 513 * sizeof(unsigned int), stream));
     launch_lstm_persistent_fwd(gx, W_h, h_all, c_all, rec,
                                reinterpret_cast<unsigned int*>(hgran),
                                abort_flag, T, B, H, stream);
@@ -254,7 +254,7 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
                               hipStream_t stream) {
   if (g_use_persistent_bwd && persistent_ok(B, H) &&
       persistent_bwd_lds(B, H) <= 160 * 1024) {
-    HIP_CHECK(hipMemsetAsync(hgran, 0, 17 * sizeof(unsigned int), stream));
+    HIP_CHECK(hipMemsetAsync(hgran, 0, 513 * sizeof(unsigned int), stream));
     launch_lstm_persistent_bwd(dY, rec, W_h_T, dG, dG_pack,
                                reinterpret_cast<unsigned int*>(hgran),
                                abort_flag, T, B, H, stream);
@@ -271,7 +271,7 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
     // dG_pack slots 0/1 alternate by step parity: a launch's phase-1
     // readers must not see its phase-2 pack writes.
     const int nbn = (H + 15) / 16;
-    unsigned int* flags = reinterpret_cast<unsigned int*>(hgran) + 32;
+    unsigned int* flags = reinterpret_cast<unsigned int*>(hgran) + 544;
     HIP_CHECK(hipMemsetAsync(flags, 0, nbn * sizeof(unsigned int), stream));
     const int64_t pstride = (int64_t)((4 * H + 31) / 32) * 2 * 64 * 8;
     launch_lstm_cell_bwd_elt(dY + (T - 1) * hstep, nullptr, dh2, dc,
@@ -322,8 +322,8 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
   auto* hg = reinterpret_cast<unsigned long long*>(hgran.data_ptr());
   auto* ab = reinterpret_cast<unsigned int*>(abort_flag.data_ptr());
   if (fused_bwd_ok(B, H) && T >= 2) {
-    // pair counters live at hgran uint32[32 ..); 2 pack slots needed
-    TORCH_CHECK(hgran.numel() * 8 >= 128 + ((H + 15) / 16) * 4,
+    // pair counters live at hgran uint32[544 ..); 2 pack slots needed
+    TORCH_CHECK(hgran.numel() * 8 >= 2176 + ((H + 15) / 16) * 4,
                 "lstm_seq_bwd: hgran too small for fused-bwd counters");
     TORCH_CHECK(dG_pack.dim() == 2 && dG_pack.size(0) >= 2,
                 "lstm_seq_bwd: fused path needs >= 2 dG_pack slots");

@@ -62,15 +62,19 @@ size_t persistent_fwd_lds(int B, int H) {
 
 #define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
 
-// pstate layout: [8] = one monotonic arrival counter (zeroed per
-// launch). Flat form: every block fetch_adds once per step and polls
-// counter >= gen*NB relaxed with s_sleep. Measured FASTER than the
-// XCD-grouped two-level variant at every grid size (4.1 vs 6.8 us at
-// 188 WGs, tools/census.hip probes) — the single hot word does not
-// congest at these poller counts.
+// pstate layout — PADDED two-level counters, every hot word on its own
+// 128-B line (zeroed per launch by the driver, words [0, 513)):
+//   word 32*g  (g=0..7): group arrival counters (blockIdx & 7)
+//   word 256:            top counter (one bump per group per step)
+//   word 288+32*g:       per-group generation release words
+// Monotonic generations, no resets. Census (tools/census.hip, 188 WGs):
+// padded two-level 1.62 us vs flat single-counter 2.82 vs two-level on
+// ADJACENT words 5.9 — the original two-level lost to flat only because
+// its 8 group counters shared one cache line, serializing every add.
 DEV_INLINE bool xcd_grid_barrier(unsigned int* pstate, int NB,
                                  unsigned int gen,
-                                 unsigned int* abort_flag) {
+                                 unsigned int* abort_flag,
+                                 bool acquire = true) {
   __shared__ int ok_s;
   // every wave drains its own write-through stores before arriving
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -78,21 +82,47 @@ DEV_INLINE bool xcd_grid_barrier(unsigned int* pstate, int NB,
   if (threadIdx.x == 0) {
     ok_s = 1;
     gu32* st = (gu32*)(uintptr_t)pstate;
-    __hip_atomic_fetch_add(&st[8], 1u, RLX_AGENT);
+    const int grp = blockIdx.x & 7;
+    const int ngroups = NB < 8 ? NB : 8;
+    const int nbg = (NB - grp + 7) / 8;  // blocks with blockIdx%8 == grp
+    unsigned int t = __hip_atomic_fetch_add(&st[grp * 32], 1u, RLX_AGENT);
+    if (t == gen * (unsigned int)nbg - 1) {
+      unsigned int tt = __hip_atomic_fetch_add(&st[256], 1u, RLX_AGENT);
+      if (tt == gen * (unsigned int)ngroups - 1)
+        for (int x = 0; x < 8; ++x)
+          __hip_atomic_store(&st[288 + x * 32], gen, RLX_AGENT);
+    }
     unsigned int spins = 0;
-    while (__hip_atomic_load(&st[8], RLX_AGENT) <
-           gen * (unsigned int)NB) {
-      __builtin_amdgcn_s_sleep(8);
+    while (__hip_atomic_load(&st[288 + grp * 32], RLX_AGENT) < gen) {
+      __builtin_amdgcn_s_sleep(2);
       if (++spins > 20000000u) {
         atomicOr(abort_flag, 1u);
         ok_s = 0;
         break;
       }
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    if (acquire) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   __syncthreads();
   return ok_s != 0;
+}
+
+// sc1 (relaxed agent-scope) 16-B load as two 8-B atomic loads: served
+// memory-side, so the freshly write-through-published h is visible with
+// NO acquire fence (and no per-step L2 invalidate). 8-B alignment holds
+// because persistent_ok requires H even.
+typedef __attribute__((address_space(1))) unsigned long long gu64;
+DEV_INLINE bf16x8 load_sc1_vec8(const bf16* p) {
+  const gu64* q = (const gu64*)(uintptr_t)p;
+  unsigned long long lo = __hip_atomic_load(q, RLX_AGENT);
+  unsigned long long hi = __hip_atomic_load(q + 1, RLX_AGENT);
+  union {
+    unsigned long long u[2];
+    bf16x8 v;
+  } r;
+  r.u[0] = lo;
+  r.u[1] = hi;
+  return r.v;
 }
 
 // paired write-through store of two adjacent bf16 values
@@ -210,7 +240,10 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     PROF_ACC(0, pt0, pt1);  // gx prefetch issue
 #endif
     if (t > 0) {
-      if (!xcd_grid_barrier(pstate, NB, (unsigned int)t, abort_flag))
+      // no acquire: h is staged below with sc1 (memory-side) loads, and
+      // nothing else read inside the loop is written cross-block
+      if (!xcd_grid_barrier(pstate, NB, (unsigned int)t, abort_flag,
+                            /*acquire=*/false))
         return;
     } else {
       __syncthreads();
@@ -231,18 +264,18 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       const bf16* hsrc = h_all + (int64_t)t * B * H;
       const int vecs = (H + 7) / 8;
       const int total = B * vecs;
-      for (int idx = t_; idx < total; idx += 8 * PCELL_THREADS) {
-        bf16x8 v[8];
+      for (int idx = t_; idx < total; idx += 16 * PCELL_THREADS) {
+        bf16x8 v[16];
         int id2 = idx;
 #pragma unroll
-        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+        for (int u = 0; u < 16; ++u, id2 += PCELL_THREADS) {
           const int ic = id2 < total ? id2 : total - 1;
           const int b = ic / vecs, k = (ic % vecs) * 8;
-          v[u] = *reinterpret_cast<const bf16x8*>(hsrc + (int64_t)b * H + k);
+          v[u] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
         }
         id2 = idx;
 #pragma unroll
-        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+        for (int u = 0; u < 16; ++u, id2 += PCELL_THREADS) {
           if (id2 < total) {
             const int b = id2 / vecs, k = (id2 % vecs) * 8;
             *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = v[u];

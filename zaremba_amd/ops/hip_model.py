@@ -81,9 +81,9 @@ class _LayerWorkspace:
         hs = _C.ext().persistent_hs(H)
         nb = (H + hs - 1) // hs
         self.rec = torch.zeros(T * nb * B * 6 * hs, dtype=bf, device=device)
-        # hgran: fwd barrier words (first 17 uint32) + fused-bwd pair
-        # counters (uint32[32 .. 32+ceil(H/16)))
-        self.hgran = torch.zeros(160, dtype=torch.int64, device=device)
+        # hgran: fwd padded-barrier words (uint32[0, 513)) + fused-bwd
+        # pair counters (uint32[544 .. 544+ceil(H/16)))
+        self.hgran = torch.zeros(768, dtype=torch.int64, device=device)
         self.abort = torch.zeros(1, dtype=torch.int32, device=device)
 
 
