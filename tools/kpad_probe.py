@@ -15,7 +15,7 @@ from zaremba_amd.models.lstm_lm import Model
 from zaremba_amd import trainer
 
 real_ext = hip_ops.ext
-counts = {"pad": 0, "nopad": 0}
+counts = {"pad": 0, "nopad": 0, "splitk": 0}
 
 
 class Proxy:
@@ -26,6 +26,10 @@ class Proxy:
         k_pad = kw.get("k_pad", a[6] if len(a) > 6 else 0)
         counts["pad" if k_pad else "nopad"] += 1
         return self._e.gemm(*a, **kw)
+
+    def gemm_splitk(self, *a, **kw):
+        counts["splitk"] += 1
+        return self._e.gemm_splitk(*a, **kw)
 
     def __getattr__(self, n):
         return getattr(self._e, n)
@@ -44,7 +48,7 @@ model.train()
 hm = model.hip()
 for it in range(13):
     if it == 3:
-        counts["pad"] = counts["nopad"] = 0
+        counts["pad"] = counts["nopad"] = counts["splitk"] = 0
         torch.cuda.synchronize()
         t0 = time.perf_counter()
     states = model.detach(states)

@@ -1,12 +1,13 @@
 // Python bindings + host-side sequence drivers for the zaremba_amd HIP
 // kernel library (gfx950).
 //
-// The per-timestep LSTM loops (forward: fused cell kernel x T; backward:
-// dgate elementwise + skinny recurrent GEMM x T) are driven from C++ and
-// cached as hipGraphs keyed on the buffer pointers/shapes — the Python
-// side allocates persistent per-layer workspaces so each training step is
-// a single graph replay per layer direction instead of 35-70 host
-// launches (the launch-overhead hot spot).
+// The per-timestep LSTM loops are driven from C++: the forward's
+// persistent kernel and the backward's fused hop+dgate step kernels
+// launch eagerly (cross-block-spin kernels hang under hipGraph replay
+// on ROCm 7.x); the per-step FALLBACK trains (fused cell x T / dgate +
+// recurrent GEMM x T) are cached as hipGraphs keyed on buffer pointers
+// so a step is one replay instead of 35-70 host launches (SURVEY.md
+// §3.5: the launch-overhead hot spot).
 #include <torch/extension.h>
 
 #include <ATen/hip/HIPContext.h>
@@ -196,7 +197,8 @@ static void lstm_seq_fwd_body(const bf16* gx, const bf16* W_h,
 
 // h_all/c_all are [T+1, B, H] with slot 0 pre-filled with (h0, c0);
 // h_pack is the zero-prefilled [T+1, KS*2*64*8] packed-h workspace;
-// hgran the [2*B*H/2] u64 granule buffer; abort a u32 flag.
+// hgran the barrier/counter state buffer (layout in _LayerWorkspace);
+// abort a u32 flag.
 static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
                          const torch::Tensor& W_pack,
                          torch::Tensor& h_all, torch::Tensor& h_pack,
@@ -206,6 +208,9 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
   int T = gx.size(0), B = gx.size(1);
   int H = h_all.size(2);
   TORCH_CHECK(gx.size(2) == 4 * H, "gx must be [T,B,4H]");
+  // padded-barrier state spans uint32 words [0, 513)
+  TORCH_CHECK(hgran.numel() * 8 >= 513 * 4,
+              "lstm_seq_fwd: hgran too small for the padded barrier");
   const bf16* gxp = bf_ptr(gx);
   const bf16* whraw = bf_ptr(W_h);
   const bf16* whp = bf_ptr(W_pack);
