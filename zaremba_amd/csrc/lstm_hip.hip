@@ -364,6 +364,24 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_packed_nt_kernel(
 typedef __attribute__((address_space(1))) unsigned int gau32;
 #define ZRLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
 
+// Optional phase census (tools/bwd_census.hip compiles this file with
+// -DZAMD_BWD_PROF); compiled out of the production .so.
+#ifdef ZAMD_BWD_PROF
+__device__ unsigned long long g_bwd_prof[512 * 8];
+#define BPROF_STAMP(v) \
+  unsigned long long v = \
+      (threadIdx.x == 0) ? __builtin_amdgcn_s_memrealtime() : 0
+#define BPROF_ACC(ph, t0, t1) \
+  if (threadIdx.x == 0) g_bwd_prof[blockIdx.x * 8 + (ph)] += (t1) - (t0)
+#else
+#define BPROF_STAMP(v) \
+  do {                 \
+  } while (0)
+#define BPROF_ACC(ph, t0, t1) \
+  do {                        \
+  } while (0)
+#endif
+
 DEV_INLINE void store_wt_f32(float* p, float v) {
   __hip_atomic_store((gau32*)(uintptr_t)p,
                      __builtin_bit_cast(unsigned int, v), ZRLX_AGENT);
@@ -406,6 +424,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   const bf16x8* pw =
       reinterpret_cast<const bf16x8*>(W_pack) + (int64_t)nb * KS * 64 + l;
 
+  BPROF_STAMP(bp0);
   f32x4 acc0 = {}, acc1 = {};
   const int nown = (ks1 - ks0 - w + 3) / 4;
   int i = 0;
@@ -433,6 +452,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
   }
 
+  BPROF_STAMP(bp1);
+  BPROF_ACC(0, bp0, bp1);  // hop MFMA (A/W loads + mfma)
   // Prefetch the dgate phase's partner-independent inputs (rec record,
   // dY, dc) NOW: the loads complete under the publish/arrive/spin that
   // follows instead of serializing into phase 2. nrows*16 <= 256, so
@@ -467,6 +488,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   }
   __syncthreads();
 
+  BPROF_STAMP(bp2);
+  BPROF_ACC(1, bp1, bp2);  // prefetch issue + red write + sync
   // publish this K-half's partial (write-through: memory-side visible
   // once the wave's vmcnt drains; the pair partner reads it sc1)
   float* mine = P + (int64_t)sk * M * N;
@@ -479,6 +502,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
+  BPROF_STAMP(bp3);
+  BPROF_ACC(2, bp2, bp3);  // partial publish + drain
   if (threadIdx.x == 0) {
     ok_s = 1;
     gau32* f = (gau32*)(uintptr_t)(flags + nb);
@@ -494,6 +519,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     }
   }
   __syncthreads();
+  BPROF_STAMP(bp4);
+  BPROF_ACC(3, bp3, bp4);  // pair arrive + spin
   if (!ok_s) return;
 
   // phase 2: dgate[t-1] for this block's half of the batch rows of its
@@ -529,6 +556,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     }
     dc[e] = dct * p_f;
   }
+  BPROF_STAMP(bp5);
+  BPROF_ACC(4, bp4, bp5);  // dgate + dG/pack stores
 }
 
 void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
