@@ -143,7 +143,7 @@ struct GraphCache {
     return cap_stream;
   }
   ~GraphCache() {
-    for (auto& kv : cache) hipGraphExecDestroy(kv.second);
+    for (auto& kv : cache) (void)hipGraphExecDestroy(kv.second);
   }
 };
 static GraphCache g_fwd_graphs, g_bwd_graphs;
@@ -369,8 +369,8 @@ static void set_use_persistent(bool v) { g_use_persistent = v; }
 static void set_use_persistent_bwd(bool v) { g_use_persistent_bwd = v; }
 static void set_use_fused_bwd(bool v) { g_use_fused_bwd = v; }
 static void clear_graphs() {
-  for (auto& kv : g_fwd_graphs.cache) hipGraphExecDestroy(kv.second);
-  for (auto& kv : g_bwd_graphs.cache) hipGraphExecDestroy(kv.second);
+  for (auto& kv : g_fwd_graphs.cache) (void)hipGraphExecDestroy(kv.second);
+  for (auto& kv : g_bwd_graphs.cache) (void)hipGraphExecDestroy(kv.second);
   g_fwd_graphs.cache.clear();
   g_bwd_graphs.cache.clear();
 }

@@ -18,7 +18,6 @@ HIP clip+SGD path with fp32 master weights + bf16 shadow rewrite.
 from __future__ import annotations
 
 import json
-import os
 import timeit
 from typing import List, Optional, Tuple
 
