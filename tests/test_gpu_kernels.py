@@ -218,9 +218,11 @@ def test_lstm_seq_fwd_persistent_matches_per_step(ext):
     assert torch.allclose(cT1, c2[T], atol=1e-6)
 
 
-def test_fused_bwd_matches_per_step_pair(ext):
-    """Fused hop+dgate backward == the per-step dgate/hop pair, bitwise
-    (same MFMA body, same f32 partial sum, same dgate math)."""
+@pytest.mark.parametrize("B", [20, 7])
+def test_fused_bwd_matches_per_step_pair(ext, B):
+    """Fused hop+dgate backward == the per-step dgate/hop pair (same MFMA
+    body, same f32 partial sum, same dgate math). B=7 exercises the odd
+    batch-row split (4 + 3 rows per pair block)."""
     from zaremba_amd.models.lstm_lm import Model
     from zaremba_amd import trainer
 
@@ -230,10 +232,10 @@ def test_fused_bwd_matches_per_step_pair(ext):
         try:
             model = Model(60, 200, 2, dropout=0.0, winit=0.05,
                           engine="hip").to(dev())
-            x = torch.randint(0, 60, (9, 20), device=dev())
-            y = torch.randint(0, 60, (9, 20), device=dev())
+            x = torch.randint(0, 60, (9, B), device=dev())
+            y = torch.randint(0, 60, (9, B), device=dev())
             model.train()
-            s = model.state_init(20)
+            s = model.state_init(B)
             scores, s = model(x, s)
             trainer.nll_loss(scores, y).backward()
             return {n: p.grad.clone() for n, p in model.named_parameters()}
