@@ -167,6 +167,8 @@ class HipModel:
             self.e.set_use_graphs(False)
         if os.environ.get("ZAREMBA_AMD_PERSISTENT", "1") == "0":
             self.e.set_use_persistent(False)
+        if os.environ.get("ZAREMBA_AMD_FUSED_BWD", "1") == "0":
+            self.e.set_use_fused_bwd(False)
         self.compute_dtype = torch.bfloat16
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,
@@ -204,6 +206,21 @@ class HipModel:
 
     def invalidate_shadows(self):
         self._shadows_fresh = False
+
+    def check_aborts(self):
+        """Raise if any grid-synchronized kernel hit its bounded-spin
+        timeout (the abort flag makes blocks EXIT instead of hanging the
+        GPU, but the step's outputs are then garbage). Called from the
+        trainer's log steps, which synchronize anyway — failure detection
+        at zero steady-state cost."""
+        for rt in self.layers:
+            if rt.ws is not None and int(rt.ws.abort.item()) != 0:
+                raise RuntimeError(
+                    "persistent/fused LSTM kernel aborted (bounded-spin "
+                    "timeout): a co-residency or synchronization failure; "
+                    "results of the affected step are invalid. Set "
+                    "ZAREMBA_AMD_PERSISTENT=0 / ZAREMBA_AMD_FUSED_BWD=0 to "
+                    "fall back to per-step kernels.")
 
     def slack_buf(self, key, shape, dtype):
         """Per-call-site cached activation buffer with zeroed storage

@@ -162,6 +162,9 @@ def train(
             # averaging (grad_scale stays 1).
             with trace_range("clip_sgd"):
                 norm = sgd_step(model, lr, max_norm)
+            if i % cadence == 0:
+                if model._hip_model is not None:
+                    model._hip_model.check_aborts()
             if i % cadence == 0 and is_rank0:
                 toc = timeit.default_timer()
                 norm_v = float(norm)
