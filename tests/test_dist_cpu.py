@@ -49,29 +49,30 @@ def _rank_main(rank, world, init_file, q):
     td.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
-def test_bucketer_allreduce_matches_serial_sum():
+@pytest.mark.timeout(180)
+@pytest.mark.parametrize("world", [2, 4])
+def test_bucketer_allreduce_matches_serial_sum(world):
     with tempfile.TemporaryDirectory() as d:
         init_file = os.path.join(d, "pg")
         ctx = mp.get_context("spawn")
         q = ctx.SimpleQueue()
-        procs = [ctx.Process(target=_rank_main, args=(r, 2, init_file, q))
-                 for r in range(2)]
+        procs = [ctx.Process(target=_rank_main, args=(r, world, init_file, q))
+                 for r in range(world)]
         for p in procs:
             p.start()
         results = {}
-        for _ in range(2):
+        for _ in range(world):
             rank, grads = q.get()
             results[rank] = grads
         for p in procs:
             p.join(60)
             assert p.exitcode == 0
 
-    # serial reference: same init, grads of batch0 + grads of batch1
+    # serial reference: same init, sum of each rank-batch's grads
     torch.manual_seed(7)
     model = Model(V, H, L, dropout=0.0, winit=0.1)
     expected = {n: torch.zeros_like(p) for n, p in model.named_parameters()}
-    for seed in (100, 101):
+    for seed in range(100, 100 + world):
         model.zero_grad()
         x, y = _make_batch(seed)
         scores, _ = model(x, model.state_init(B))
@@ -80,6 +81,7 @@ def test_bucketer_allreduce_matches_serial_sum():
             expected[n] += p.grad
     for n in expected:
         r0 = torch.from_numpy(results[0][n])
-        r1 = torch.from_numpy(results[1][n])
         assert torch.allclose(r0, expected[n], atol=1e-5), n
-        assert torch.allclose(r0, r1, atol=1e-7), n
+        for r in range(1, world):
+            assert torch.allclose(
+                r0, torch.from_numpy(results[r][n]), atol=1e-7), n
