@@ -3,8 +3,11 @@
 // Covers the framework's dense-GEMM call sites:
 //   K2  input-side gate GEMM   gx = x @ W_x^T        (NT, reference model.py:35)
 //   K6  output projection      scores = h @ fc.W^T   (NT, reference model.py:67)
-//   K8  backward data GEMMs    dx = dG @ W (NT via pre-transposed shadow
-//       weights) and weight-grad GEMMs dW = dG^T @ x (TN).
+//   K8  backward data GEMMs    dx = dG @ W (NT via pre-transposed,
+//       K-zero-padded shadow weights; 2-way split-K on the ~1-block/CU
+//       shapes) and weight-grad GEMMs dW = dG^T @ x (as transpose + NT
+//       over K-padded temps; the TN staging path exists but is
+//       register-starved and unused on the hot path).
 //
 // Design notes (measured on MI355X, see profiles/):
 //   * BMxBN output tile (128x128 or 64x64 chosen by grid size so skinny
