@@ -127,3 +127,51 @@ def test_ensemble_perplexity_beats_worst_member():
         ppls.append(trainer.perplexity(ds, m, batch_size=4))
     ens = ensemble_perplexity(ds, models, batch_size=4)
     assert ens <= max(ppls) + 1e-6
+
+
+def test_checkpoint_rejects_wrong_version(tmp_path):
+    torch.manual_seed(3)
+    ds, vocab = _tiny_data()
+    model = Model(vocab, 8, 1, dropout=0.0, winit=0.1)
+    path = str(tmp_path / "ck.pt")
+    save_checkpoint(path, model, epoch=1, lr=0.5)
+    payload = torch.load(path, weights_only=False)
+    payload["format_version"] = 999
+    torch.save(payload, path)
+    with pytest.raises(ValueError):
+        build_model_from_checkpoint(path)
+
+
+def test_jsonl_mirror_content(tmp_path):
+    """--jsonl writes one machine-readable record per log event with the
+    same quantities the console line carries."""
+    import json
+
+    torch.manual_seed(4)
+    ds, vocab = _tiny_data()
+    model = Model(vocab, 8, 1, dropout=0.0, winit=0.1)
+    path = str(tmp_path / "log.jsonl")
+    trainer.train((ds, ds, ds), model, epochs=3, epoch_threshold=1, lr=1.0,
+                  factor=2.0, max_norm=5.0, batch_size=4, jsonl_path=path)
+    events = [json.loads(l) for l in open(path)]
+    kinds = [e["event"] for e in events]
+    assert "step" in kinds and "epoch" in kinds and "final" in kinds
+    step = next(e for e in events if e["event"] == "step")
+    for f in ("train_loss", "wps", "grad_norm", "lr", "elapsed_s"):
+        assert f in step, f
+    epochs = [e for e in events if e["event"] == "epoch"]
+    assert len(epochs) == 3 and all("valid_ppl" in e for e in epochs)
+    # LR decay off-by-one (reference main.py:105-106): decay only for
+    # epoch index > factor_epoch, so epochs 0,1 run at lr and epoch 2
+    # at lr/factor
+    assert epochs[0]["lr"] == epochs[1]["lr"] == pytest.approx(1.0)
+    assert epochs[2]["lr"] == pytest.approx(0.5)
+    final = next(e for e in events if e["event"] == "final")
+    assert "test_ppl" in final
+
+
+def test_model_ctor_validation():
+    with pytest.raises(ValueError):
+        Model(10, 8, 1, dropout=0.0, winit=0.1, lstm_type="nonsense")
+    with pytest.raises(ValueError):
+        Model(10, 8, 1, dropout=0.0, winit=0.1, engine="nonsense")
