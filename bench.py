@@ -55,6 +55,15 @@ def main():
         os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
     rank, world = zdist.rank(), zdist.world_size()
+    if args.gpus != world:
+        # --gpus is authoritative: a mismatch means the launch was wrong
+        # (e.g. `bench.py --gpus 8` run single-process would silently
+        # bench dp1). Fail loudly instead of reporting the wrong config.
+        raise SystemExit(
+            f"bench.py: --gpus {args.gpus} but torch.distributed world size "
+            f"is {world}. Launch with `python -m torch.distributed.run "
+            f"--nnodes=1 --nproc-per-node {args.gpus} --master-addr "
+            f"127.0.0.1 bench.py --gpus {args.gpus} ...` (one rank per GPU).")
     has_gpu = torch.cuda.is_available()
     device = torch.device("cuda", zdist.local_rank()) if has_gpu else \
         torch.device("cpu")

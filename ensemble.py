@@ -10,8 +10,10 @@ reference kept them only in memory).
 
 Multi-GPU (launch via torchrun, one rank per GPU): models are trained in
 parallel round-robin across ranks (model i on rank i % world), each saved
-to --save_dir; rank 0 then reports the incremental ensemble perplexities
-over the saved checkpoints.
+to --save_dir and kept resident on its trainer rank; the incremental
+ensemble perplexities are then evaluated DISTRIBUTED — every rank forwards
+its own members and one RCCL all-reduce(SUM) per batch merges the
+probability sums (reference math ensemble.py:97-126; BASELINE config 5).
 """
 
 import argparse
@@ -21,8 +23,9 @@ import torch
 
 from zaremba_amd import data as zdata
 from zaremba_amd import trainer
-from zaremba_amd.checkpoint import build_model_from_checkpoint, save_checkpoint
-from zaremba_amd.ensemble_eval import ensemble_perplexity
+from zaremba_amd.checkpoint import save_checkpoint
+from zaremba_amd.ensemble_eval import (ensemble_perplexity,
+                                       ensemble_perplexity_distributed)
 from zaremba_amd.models.lstm_lm import Model
 from zaremba_amd.parallel import dist as zdist
 
@@ -85,12 +88,15 @@ def load_data(args):
     return _ld(args)
 
 
-def train_one(args, vocab_size, data, model_num):
+def train_one(args, vocab_size, data, model_num, jsonl_path=None):
     if args.seed is not None:
         torch.manual_seed(args.seed + 1000 * model_num)
     engine = args.engine
-    if args.device.type == "cuda" and args.dtype == "fp32" \
-            and engine == "auto":
+    if args.device.type == "cuda" and args.dtype == "fp32":
+        if engine == "hip":
+            raise SystemExit(
+                "--dtype fp32 is not supported by --engine hip (bf16 "
+                "compute / fp32 masters); use --engine eager or auto.")
         engine = "eager"
         os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     model = Model(vocab_size, args.hidden_size, args.layer_num, args.dropout,
@@ -101,12 +107,28 @@ def train_one(args, vocab_size, data, model_num):
     trainer.train(data, model, args.total_epochs, args.factor_epoch,
                   args.learning_rate, args.factor, args.max_grad_norm,
                   args.batch_size, log_every=800, model_num=model_num,
-                  jsonl_path=args.jsonl, is_rank0=True)
+                  jsonl_path=jsonl_path, is_rank0=True)
     return model
+
+
+def jsonl_for_model(args, model_num):
+    """Per-model JSONL path. In multi-rank mode every rank logs with
+    is_rank0=True, so a shared file would interleave concurrent appends
+    (advisor finding); suffix with the model number instead."""
+    if args.jsonl is None:
+        return None
+    if zdist.world_size() == 1:
+        return args.jsonl
+    root, ext = os.path.splitext(args.jsonl)
+    return f"{root}.model{model_num}{ext or '.jsonl'}"
 
 
 def main():
     args = build_parser().parse_args()
+    if args.engine == "hip" and args.dtype == "fp32":
+        raise SystemExit(
+            "--dtype fp32 is not supported by --engine hip (bf16 compute / "
+            "fp32 masters); use --engine eager or auto.")
     if args.engine == "eager":
         os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
@@ -128,35 +150,41 @@ def main():
             raise SystemExit("--save_dir is required for multi-rank ensemble "
                              "training")
         os.makedirs(args.save_dir, exist_ok=True)
-        # Parallel: model i trains on rank i % world (diversity = seeds/init).
+        # Parallel: model i trains on rank i % world (diversity = seeds/init)
+        # and STAYS resident on its trainer rank for the distributed eval.
+        my_models = {}
         for i in range(args.ensemble_num):
             if i % world == zdist.rank():
-                model = train_one(args, vocab_size, data, i + 1)
+                model = train_one(args, vocab_size, data, i + 1,
+                                  jsonl_path=jsonl_for_model(args, i + 1))
                 save_checkpoint(os.path.join(args.save_dir, f"model_{i + 1}.pt"),
                                 model, epoch=args.total_epochs,
                                 lr=args.learning_rate)
-                del model
+                my_models[i + 1] = model
         zdist.barrier()
-        if zdist.is_rank0():
-            models = {}
-            for i in range(args.ensemble_num):
-                path = os.path.join(args.save_dir, f"model_{i + 1}.pt")
-                m, _ = build_model_from_checkpoint(path, engine=args.engine)
-                m.to(args.device)
-                models[f"model {i + 1}"] = m
-                val_perp = ensemble_perplexity(vld, models, args.batch_size)
+        # Distributed incremental eval (BASELINE config 5 / SURVEY K13):
+        # for each ensemble size k, every rank forwards only the members it
+        # trained; one all-reduce(SUM) per batch merges the prob sums over
+        # RCCL/xGMI. Ranks owning no member <= k contribute zeros.
+        for k in range(1, args.ensemble_num + 1):
+            subset = {f"model {j}": m for j, m in my_models.items() if j <= k}
+            val_perp = ensemble_perplexity_distributed(
+                vld, subset, k, args.batch_size, vocab_size, args.device)
+            tst_perp = ensemble_perplexity_distributed(
+                tst, subset, k, args.batch_size, vocab_size, args.device)
+            if zdist.is_rank0():
                 print("Validation set perplexity of {} averaged models: {:.3f}"
-                      .format(i + 1, val_perp))
-                tst_perp = ensemble_perplexity(tst, models, args.batch_size)
+                      .format(k, val_perp))
                 print("Test set perplexity of {} averaged models: {:.3f}\n"
-                      .format(i + 1, tst_perp))
+                      .format(k, tst_perp))
         zdist.finalize()
         return
 
     # Single process: the reference's sequential flow (ensemble.py:166-182).
     models = {}
     for i in range(args.ensemble_num):
-        model = train_one(args, vocab_size, data, i + 1)
+        model = train_one(args, vocab_size, data, i + 1,
+                          jsonl_path=jsonl_for_model(args, i + 1))
         models["model {:d}".format(i + 1)] = model
         if args.save_dir:
             os.makedirs(args.save_dir, exist_ok=True)

@@ -118,6 +118,13 @@ def load_data(args):
 
 def main():
     args = build_parser().parse_args()
+    if args.engine == "hip" and args.dtype == "fp32":
+        # The HIP kernel path is bf16-compute / fp32-master by design;
+        # refuse up front rather than silently computing in bf16.
+        raise SystemExit(
+            "--dtype fp32 is not supported by --engine hip (the HIP "
+            "kernels compute in bf16 with fp32 master weights). Use "
+            "--engine eager (or auto) for full-fp32 compute.")
     if args.engine == "eager":
         os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     zdist.maybe_init()
@@ -138,19 +145,22 @@ def main():
 
     start_epoch, start_lr = 0, args.learning_rate
     if args.resume and os.path.exists(args.resume):
-        from zaremba_amd.checkpoint import build_model_from_checkpoint
+        from zaremba_amd.checkpoint import (build_model_from_checkpoint,
+                                            restore_rng)
         model, payload = build_model_from_checkpoint(args.resume, engine=args.engine)
         start_epoch = payload["epoch"]
+        # Restore the RNG streams so a resumed run's dropout masks and
+        # any data draws continue the unbroken run's sequence.
+        restore_rng(payload)
         if zdist.is_rank0():
             print(f"Resumed from {args.resume} at epoch {start_epoch}.")
     else:
         model = Model(vocab_size, args.hidden_size, args.layer_num,
                       args.dropout, args.winit, args.lstm_type,
                       engine=args.engine)
-    if args.device.type == "cuda" and args.dtype == "fp32" \
-            and args.engine == "auto":
-        # fp32 compute on GPU = the eager engine (the HIP kernel path is
-        # bf16-compute / fp32-master by design)
+    if args.device.type == "cuda" and args.dtype == "fp32":
+        # engine auto (hip+fp32 was rejected at parse time): fp32 compute
+        # on GPU = the eager engine
         model.engine = "eager"
         os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     model.to(args.device)

@@ -60,3 +60,24 @@ def test_main_cli_missing_ptb_message(tmp_path):
     r = run_cli(["main.py", "--device", "cpu", "--data_dir", str(tmp_path)])
     assert r.returncode != 0
     assert "synthetic" in (r.stderr + r.stdout)
+
+
+@pytest.mark.timeout(120)
+def test_fp32_with_hip_engine_rejected():
+    """--dtype fp32 --engine hip must fail loudly (the HIP path computes
+    in bf16 with fp32 masters); round-1 bug: it silently ran bf16."""
+    for script in ("main.py", "ensemble.py"):
+        r = run_cli([script, "--device", "cpu", "--engine", "hip",
+                     "--dtype", "fp32", "--data", "synthetic:vocab=20,tokens=500"])
+        assert r.returncode != 0, script
+        assert "fp32 is not supported" in (r.stderr + r.stdout), script
+
+
+@pytest.mark.timeout(120)
+def test_bench_gpus_flag_is_authoritative():
+    """bench.py --gpus N run single-process (world=1) must fail loudly
+    instead of silently benchmarking dp1 (protects the driver's scaling
+    runs)."""
+    r = run_cli(["bench.py", "--gpus", "2", "--steps", "1", "--warmup", "0"])
+    assert r.returncode != 0
+    assert "world size" in (r.stderr + r.stdout)

@@ -85,3 +85,74 @@ def test_bucketer_allreduce_matches_serial_sum(world):
         for r in range(1, world):
             assert torch.allclose(
                 r0, torch.from_numpy(results[r][n]), atol=1e-7), n
+
+
+# ---------------------------------------------------------------------------
+# Distributed ensemble averaging (K13 / BASELINE config 5)
+# ---------------------------------------------------------------------------
+
+K_MODELS = 3
+
+
+def _build_member(num):
+    torch.manual_seed(1000 * num)
+    return Model(V, H, 1, dropout=0.0, winit=0.1)
+
+
+def _eval_data(n_batches=3):
+    out = []
+    for s in range(n_batches):
+        out.append(_make_batch(500 + s))
+    return out
+
+
+def _ens_rank_main(rank, world, init_file, q):
+    td.init_process_group("gloo", init_method=f"file://{init_file}",
+                          rank=rank, world_size=world)
+    from zaremba_amd.ensemble_eval import ensemble_perplexity_distributed
+    # member i (1-based) lives on rank (i-1) % world — rank 1 owns none
+    # when K_MODELS < world leaves a gap; the zero-contribution path is
+    # exercised by the k=1 increment (only rank 0 owns model 1).
+    my = {f"model {i + 1}": _build_member(i + 1)
+          for i in range(K_MODELS) if i % world == rank}
+    data = _eval_data()
+    ppls = []
+    for k in range(1, K_MODELS + 1):
+        subset = {name: m for name, m in my.items()
+                  if int(name.split()[1]) <= k}
+        ppls.append(ensemble_perplexity_distributed(
+            data, subset, k, B, V, torch.device("cpu")))
+    q.put((rank, ppls))
+    td.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_ensemble_eval_matches_sequential():
+    """Per-rank prob sums + all-reduce(SUM)/k == the sequential
+    all-models-on-one-process averaging (reference ensemble.py:97-126),
+    for every incremental ensemble size k."""
+    world = 2
+    with tempfile.TemporaryDirectory() as d:
+        init_file = os.path.join(d, "pg")
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        procs = [ctx.Process(target=_ens_rank_main,
+                             args=(r, world, init_file, q))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(world):
+            rank, ppls = q.get()
+            results[rank] = ppls
+        for p in procs:
+            p.join(60)
+            assert p.exitcode == 0
+
+    from zaremba_amd.ensemble_eval import ensemble_perplexity
+    data = _eval_data()
+    for k in range(1, K_MODELS + 1):
+        models = {f"model {i + 1}": _build_member(i + 1) for i in range(k)}
+        expected = ensemble_perplexity(data, models, B)
+        assert results[0][k - 1] == pytest.approx(expected, rel=1e-6), k
+        assert results[1][k - 1] == pytest.approx(expected, rel=1e-6), k

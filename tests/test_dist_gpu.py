@@ -1,0 +1,120 @@
+"""Multi-rank data parallelism over the REAL RCCL backend, on a single
+GPU (pytest -m gpu).
+
+Round-1 gap: the entire DP story rested on CPU gloo. Here two ranks share
+device 0 (the ZAREMBA_AMD_ONE_GPU escape in parallel/dist.py exists for
+exactly this) and run the full HIP-engine step — fused backward into the
+bucketer's flat grad views, async RCCL all-reduce per bucket — asserting
+multi-rank grad equality against the single-process sum of the same two
+per-rank batches. This is the correctness contract the driver's 8-GPU
+scaling run relies on, proven on RCCL itself rather than gloo.
+"""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+pytestmark = pytest.mark.gpu
+
+V, H, L, B, T = 64, 128, 2, 8, 6
+
+
+def _make_batch(seed):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randint(0, V, (T, B), generator=g)
+    y = torch.randint(0, V, (T, B), generator=g)
+    return x, y
+
+
+def _grads_for_batch(model, seed, dev):
+    from zaremba_amd import trainer
+    x, y = _make_batch(seed)
+    states = model.state_init(B)
+    scores, _ = model(x.to(dev), states)
+    loss = trainer.nll_loss(scores, y.to(dev))
+    loss.backward()
+
+
+def _rank_main(rank, world, init_file, q):
+    try:
+        os.environ["ZAREMBA_AMD_ONE_GPU"] = "1"
+        os.environ["ZAREMBA_AMD_ALLREDUCE_FP32"] = "1"  # exact sum
+        import torch.distributed as td
+        torch.cuda.set_device(0)
+        td.init_process_group("nccl", init_method=f"file://{init_file}",
+                              rank=rank, world_size=world)
+        from zaremba_amd.models.lstm_lm import Model
+        from zaremba_amd.parallel.bucketer import GradBucketer
+        dev = torch.device("cuda", 0)
+        torch.manual_seed(7)
+        model = Model(V, H, L, dropout=0.0, winit=0.1, engine="hip").to(dev)
+        dp = GradBucketer(model, bucket_bytes=65536)  # several buckets
+        dp.zero_grad()
+        model.train()
+        _grads_for_batch(model, 100 + rank, dev)
+        dp.finalize_backward()
+        torch.cuda.synchronize()
+        grads = {n: p.grad.cpu().numpy().copy()
+                 for n, p in model.named_parameters()}
+        q.put((rank, grads))
+        td.destroy_process_group()
+    except Exception as e:  # ship the failure instead of hanging the queue
+        q.put((rank, f"ERROR: {type(e).__name__}: {e}"))
+        raise
+
+
+@pytest.mark.timeout(420)
+def test_rccl_bucketer_allreduce_two_ranks_one_gpu():
+    world = 2
+    ctx = mp.get_context("spawn")
+    with tempfile.TemporaryDirectory() as d:
+        init_file = os.path.join(d, "pg")
+        q = ctx.SimpleQueue()
+        procs = [ctx.Process(target=_rank_main,
+                             args=(r, world, init_file, q))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        results = {}
+        try:
+            for _ in range(world):
+                rank, grads = q.get()
+                if isinstance(grads, str):
+                    pytest.fail(f"rank {rank}: {grads}")
+                results[rank] = grads
+        finally:
+            for p in procs:
+                p.join(120)
+                if p.is_alive():
+                    p.terminate()
+    for p in procs:
+        assert p.exitcode == 0
+
+    # single-process reference: same init, sum of both per-rank batch grads
+    from zaremba_amd.models.lstm_lm import Model
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(7)
+    model = Model(V, H, L, dropout=0.0, winit=0.1, engine="hip").to(dev)
+    model.train()
+    expected = {n: torch.zeros_like(p) for n, p in model.named_parameters()}
+    for seed in (100, 101):
+        model.zero_grad(set_to_none=False)
+        _grads_for_batch(model, seed, dev)
+        for n, p in model.named_parameters():
+            expected[n] += p.grad
+    torch.cuda.synchronize()
+    for n, e in expected.items():
+        scale = e.abs().max().item() + 1e-6
+        r0 = torch.from_numpy(results[0][n]).to(dev)
+        r1 = torch.from_numpy(results[1][n]).to(dev)
+        # ranks agree bitwise (same all-reduce result)
+        assert torch.equal(r0, r1), n
+        # vs serial sum: identical kernels/batches; embedding-bwd atomics
+        # reorder fp32 adds, hence the small tolerance
+        err = (r0 - e).abs().max().item() / scale
+        assert err < 1e-3, (n, err)
+
+

@@ -356,6 +356,25 @@ def test_lsm_nll_fwd_bwd_vs_eager(ext):
     assert (g1 - scores2.grad).abs().max().item() < 1e-5
 
 
+def test_softmax_acc_vs_eager(ext):
+    """K13 fused ensemble accumulate: acc += softmax(scores) rowwise."""
+    torch.manual_seed(11)
+    N, V = 700, 10000
+    acc = torch.rand(N, V, device=dev(), dtype=torch.float32)
+    expected = acc.clone()
+    for k in range(3):
+        scores = torch.randn(N, V, device=dev()) * (3 + k)
+        expected += torch.softmax(scores, dim=1)
+        ext.softmax_acc(scores, acc)
+    assert (acc - expected).abs().max().item() < 1e-4
+    # odd shape / tail handling
+    acc2 = torch.zeros(13, 1037, device=dev())
+    s2 = torch.randn(13, 1037, device=dev()) * 8
+    ext.softmax_acc(s2, acc2)
+    assert (acc2 - torch.softmax(s2, dim=1)).abs().max().item() < 1e-5
+    assert (acc2.sum(dim=1) - 1.0).abs().max().item() < 1e-4
+
+
 # ---------------------------------------------------------------------------
 # SGD
 # ---------------------------------------------------------------------------

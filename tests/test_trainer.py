@@ -170,6 +170,29 @@ def test_jsonl_mirror_content(tmp_path):
     assert "test_ppl" in final
 
 
+def test_resume_restores_rng_state(tmp_path):
+    """restore_rng puts the torch RNG back where the checkpoint left it,
+    so a resumed run's dropout/data draws continue the unbroken run's
+    sequence (round-1 gap: saved but never restored)."""
+    from zaremba_amd.checkpoint import load_checkpoint, restore_rng
+
+    torch.manual_seed(5)
+    model = Model(10, 8, 1, dropout=0.0, winit=0.1)
+    torch.rand(100)  # advance the stream to a nontrivial point
+    expected_next = None
+    path = str(tmp_path / "rng.pt")
+    state_at_save = torch.get_rng_state().clone()
+    save_checkpoint(path, model, epoch=1, lr=1.0)
+    expected_next = torch.rand(8)
+    # perturb the stream, then restore from the checkpoint
+    torch.manual_seed(999)
+    torch.rand(13)
+    payload = load_checkpoint(path)
+    restore_rng(payload)
+    assert torch.equal(torch.get_rng_state(), state_at_save)
+    assert torch.equal(torch.rand(8), expected_next)
+
+
 def test_model_ctor_validation():
     with pytest.raises(ValueError):
         Model(10, 8, 1, dropout=0.0, winit=0.1, lstm_type="nonsense")

@@ -38,6 +38,11 @@ def save_checkpoint(path: str, model, epoch: int = 0, lr: float = 0.0,
         "epoch": epoch,
         "lr": lr,
         "torch_rng_state": torch.get_rng_state(),
+        # All visible device generators (the HIP dropout kernel draws its
+        # philox seed from python's secrets, re-seeded per process, but
+        # eager-path dropout and any torch.cuda sampling use these).
+        "cuda_rng_state": (torch.cuda.get_rng_state_all()
+                           if torch.cuda.is_available() else None),
     }
     if extra:
         payload.update(extra)
@@ -51,6 +56,22 @@ def load_checkpoint(path: str, map_location="cpu"):
             f"checkpoint {path}: unsupported format_version "
             f"{payload.get('format_version')!r}")
     return payload
+
+
+def restore_rng(payload: dict):
+    """Restore the RNG streams a checkpoint captured, so a resumed run's
+    dropout masks / data draws continue the unbroken run's sequence
+    (round-1 gap: the state was saved but never restored)."""
+    state = payload.get("torch_rng_state")
+    if state is not None:
+        torch.set_rng_state(state.cpu() if torch.is_tensor(state) else
+                            torch.as_tensor(state, dtype=torch.uint8))
+    cuda_state = payload.get("cuda_rng_state")
+    if cuda_state is not None and torch.cuda.is_available():
+        # Tolerate resuming on a box with fewer devices than the saver.
+        n = min(len(cuda_state), torch.cuda.device_count())
+        for i in range(n):
+            torch.cuda.set_rng_state(cuda_state[i], i)
 
 
 def build_model_from_checkpoint(path: str, engine: str = "auto"):

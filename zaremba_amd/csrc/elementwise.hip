@@ -241,6 +241,71 @@ void launch_lsm_nll_bwd(const float* scores, const float* lse,
 }
 
 // ---------------------------------------------------------------------------
+// K13: fused softmax-accumulate for ensemble probability averaging
+// (reference ensemble.py:100-105). acc[n,:] += softmax(scores[n,:]) in one
+// launch per member — no per-model probability tensors, no k-way stack.
+// Pass 1 per row: online (max, sumexp) as in lsm_nll_fwd; pass 2 re-reads
+// the row (L2-hot, V=10k -> 40 KB) and accumulates exp(s - M) / A.
+// ---------------------------------------------------------------------------
+__global__ void softmax_acc_kernel(const float* __restrict__ scores,
+                                   float* __restrict__ acc, int N, int V) {
+  __shared__ float sm[8], ss[8];
+  __shared__ float row_m, row_a;
+  int row = blockIdx.x;
+  if (row >= N) return;
+  const float* s = scores + (int64_t)row * V;
+  float m = -INFINITY, a = 0.f;
+  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+    float x = s[v];
+    if (x > m) {
+      a = a * __expf(m - x) + 1.f;
+      m = x;
+    } else {
+      a += __expf(x - m);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float m2 = __shfl_down(m, off, 64);
+    float a2 = __shfl_down(a, off, 64);
+    float mn = fmaxf(m, m2);
+    float pa = (m == -INFINITY) ? 0.f : a * __expf(m - mn);
+    float pb = (m2 == -INFINITY) ? 0.f : a2 * __expf(m2 - mn);
+    a = pa + pb;
+    m = mn;
+  }
+  const int nw = blockDim.x / 64;
+  if (lane_id() == 0) {
+    sm[wave_id()] = m;
+    ss[wave_id()] = a;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float M = sm[0], A = ss[0];
+    for (int w = 1; w < nw; ++w) {
+      float mn = fmaxf(M, sm[w]);
+      float pa = (M == -INFINITY) ? 0.f : A * __expf(M - mn);
+      float pb = (sm[w] == -INFINITY) ? 0.f : ss[w] * __expf(sm[w] - mn);
+      A = pa + pb;
+      M = mn;
+    }
+    row_m = M;
+    row_a = A;
+  }
+  __syncthreads();
+  const float M = row_m, invA = 1.f / row_a;
+  float* d = acc + (int64_t)row * V;
+  for (int v = threadIdx.x; v < V; v += blockDim.x)
+    d[v] += __expf(s[v] - M) * invA;
+}
+
+void launch_softmax_acc(const float* scores, float* acc, int N, int V,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(softmax_acc_kernel, dim3(N), dim3(256), 0, stream,
+                     scores, acc, N, V);
+}
+
+// ---------------------------------------------------------------------------
 // K9/K10: fused grad clip + SGD
 // ---------------------------------------------------------------------------
 __global__ void norm2_accum_kernel(const float* __restrict__ g, int64_t n,

@@ -230,9 +230,13 @@ static void lstm_seq_fwd(const torch::Tensor& gx, const torch::Tensor& W_h,
                       stream);
     return;
   }
+  // EVERY device pointer the captured launches bake in must key the
+  // graph (advisor finding: an omitted pointer reallocated alongside
+  // reused keyed ones would replay a stale graph into freed memory).
   std::vector<uintptr_t> key{(uintptr_t)gxp, (uintptr_t)whraw, (uintptr_t)whp,
                              (uintptr_t)hp, (uintptr_t)hpk, (uintptr_t)cp,
-                             (uintptr_t)gp, (uintptr_t)hg, (uintptr_t)T,
+                             (uintptr_t)gp, (uintptr_t)rp, (uintptr_t)hg,
+                             (uintptr_t)ab, (uintptr_t)T,
                              (uintptr_t)B, (uintptr_t)H};
   auto it = g_fwd_graphs.cache.find(key);
   if (it == g_fwd_graphs.cache.end()) {
@@ -344,9 +348,12 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
                       ab, T, B, H, stream);
     return;
   }
+  // Complete pointer key (see the fwd-graph key comment).
   std::vector<uintptr_t> key{(uintptr_t)dyp, (uintptr_t)gp, (uintptr_t)rp,
+                             (uintptr_t)cp, (uintptr_t)whtp,
                              (uintptr_t)wtp, (uintptr_t)dgp, (uintptr_t)dgpk,
-                             (uintptr_t)dhp, (uintptr_t)dcp, (uintptr_t)T,
+                             (uintptr_t)dhp, (uintptr_t)dcp, (uintptr_t)hg,
+                             (uintptr_t)ab, (uintptr_t)T,
                              (uintptr_t)B, (uintptr_t)H};
   auto it = g_bwd_graphs.cache.find(key);
   if (it == g_bwd_graphs.cache.end()) {
@@ -516,6 +523,15 @@ static void colsum_f32(const torch::Tensor& in, torch::Tensor& out) {
                     current_stream());
 }
 
+// K13: acc += softmax(scores) rowwise (ensemble probability averaging,
+// reference ensemble.py:100-105); scores/acc f32 [N, V].
+static void softmax_acc(const torch::Tensor& scores, torch::Tensor& acc) {
+  int N = scores.size(0), V = scores.size(1);
+  TORCH_CHECK(acc.size(0) == N && acc.size(1) == V,
+              "softmax_acc: acc must match scores' shape");
+  launch_softmax_acc(f_ptr(scores), f_ptr_mut(acc), N, V, current_stream());
+}
+
 // dst is [C, ldd] with ldd >= R; columns [R, ldd) are left untouched
 // (pre-zero them once when using the pad as a GEMM K extension).
 static void transpose_bf16(const torch::Tensor& src, torch::Tensor& dst) {
@@ -555,6 +571,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose_bf16", &zamd::transpose_bf16);
   m.def("colsum_bf16", &zamd::colsum_bf16);
   m.def("colsum_f32", &zamd::colsum_f32);
+  m.def("softmax_acc", &zamd::softmax_acc);
   m.def("set_use_graphs", &zamd::set_use_graphs);
   m.def("set_use_persistent", &zamd::set_use_persistent);
   m.def("persistent_hs", &zamd::persistent_hs);

@@ -96,5 +96,7 @@ void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                         hipStream_t stream);
 void launch_colsum_f32(const float* in, float* out, int R, int C,
                        hipStream_t stream);
+void launch_softmax_acc(const float* scores, float* acc, int N, int V,
+                        hipStream_t stream);
 
 }  // namespace zamd

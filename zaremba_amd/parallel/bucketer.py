@@ -11,7 +11,11 @@ Design (MI355X/xGMI-first, SURVEY.md §5 'Distributed communication'):
   * the per-param post-accumulate-grad hook counts the bucket down and
     fires an async RCCL all-reduce (SUM) the moment the bucket completes;
     ``finalize_backward()`` joins all in-flight works before the fused
-    clip+SGD consumes the grads (which divides by world_size),
+    clip+SGD consumes the grads. Grads stay SUMMED across ranks — no
+    division by world_size anywhere: the reference loss sums over the
+    batch dimension, so summed grads are exactly the reference's
+    gradients at batch_size = world * B (trainer.py's grad_scale=1
+    comment; do NOT "fix" this to averaging),
   * xGMI is point-to-point (7 links/GPU): grads for the Large model are
     only ~266 MB fp32, so latency and overlap dominate — the default
     bucket size (25 MB) keeps a handful of in-flight collectives without
@@ -63,6 +67,12 @@ class GradBucketer:
                           or not torch.cuda.is_available()
                           else torch.bfloat16)
         self.world_size = td.get_world_size()
+        if td.get_rank() == 0:
+            # The wire dtype is a numerics choice (bf16 halves xGMI bytes
+            # but rounds the summands); make it visible in every DP log.
+            print(f"[zaremba_amd] DP gradient all-reduce wire dtype: "
+                  f"{str(comm_dtype).replace('torch.', '')} "
+                  f"(ZAREMBA_AMD_ALLREDUCE_FP32=1 forces fp32)")
         params = [p for p in model.parameters() if p.requires_grad]
         params.reverse()  # backward completion order
         self.buckets: List[_Bucket] = []
