@@ -1,13 +1,17 @@
-"""Multi-rank data parallelism over the REAL RCCL backend, on a single
-GPU (pytest -m gpu).
+"""Multi-rank data parallelism over the REAL RCCL backend (pytest -m gpu).
 
-Round-1 gap: the entire DP story rested on CPU gloo. Here two ranks share
-device 0 (the ZAREMBA_AMD_ONE_GPU escape in parallel/dist.py exists for
-exactly this) and run the full HIP-engine step — fused backward into the
-bucketer's flat grad views, async RCCL all-reduce per bucket — asserting
-multi-rank grad equality against the single-process sum of the same two
-per-rank batches. This is the correctness contract the driver's 8-GPU
-scaling run relies on, proven on RCCL itself rather than gloo.
+Round-1 gap: the entire DP story rested on CPU gloo. Here two ranks run
+the full HIP-engine step — fused backward into the bucketer's flat grad
+views, async RCCL all-reduce per bucket — asserting multi-rank grad
+equality against the single-process sum of the same two per-rank
+batches. This is the correctness contract the driver's 8-GPU scaling
+run relies on, proven on RCCL itself rather than gloo.
+
+Needs >= 2 visible devices: RCCL (like NCCL) refuses two ranks on one
+device ("Duplicate GPU detected", measured on RCCL 2.26 — so the
+ZAREMBA_AMD_ONE_GPU route cannot exercise collectives). On a 1-GPU box
+the MI355X can be split into CPX compute partitions (amd-smi) to make
+this test run for real — see tools/gpu_cpx_dp.sh.
 """
 
 import os
@@ -40,15 +44,14 @@ def _grads_for_batch(model, seed, dev):
 
 def _rank_main(rank, world, init_file, q):
     try:
-        os.environ["ZAREMBA_AMD_ONE_GPU"] = "1"
         os.environ["ZAREMBA_AMD_ALLREDUCE_FP32"] = "1"  # exact sum
         import torch.distributed as td
-        torch.cuda.set_device(0)
+        torch.cuda.set_device(rank)
         td.init_process_group("nccl", init_method=f"file://{init_file}",
                               rank=rank, world_size=world)
         from zaremba_amd.models.lstm_lm import Model
         from zaremba_amd.parallel.bucketer import GradBucketer
-        dev = torch.device("cuda", 0)
+        dev = torch.device("cuda", rank)
         torch.manual_seed(7)
         model = Model(V, H, L, dropout=0.0, winit=0.1, engine="hip").to(dev)
         dp = GradBucketer(model, bucket_bytes=65536)  # several buckets
@@ -67,7 +70,11 @@ def _rank_main(rank, world, init_file, q):
 
 
 @pytest.mark.timeout(420)
-def test_rccl_bucketer_allreduce_two_ranks_one_gpu():
+def test_rccl_bucketer_allreduce_two_ranks():
+    if torch.cuda.device_count() < 2:
+        pytest.skip("RCCL needs one device per rank (duplicate-GPU refusal "
+                    "verified on RCCL 2.26); run on a multi-GPU box or CPX-"
+                    "partitioned MI355X (tools/gpu_cpx_dp.sh)")
     world = 2
     ctx = mp.get_context("spawn")
     with tempfile.TemporaryDirectory() as d:

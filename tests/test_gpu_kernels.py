@@ -218,17 +218,20 @@ def test_lstm_seq_fwd_persistent_matches_per_step(ext):
     assert torch.allclose(cT1, c2[T], atol=1e-6)
 
 
-@pytest.mark.parametrize("B", [20, 7])
-def test_fused_bwd_matches_per_step_pair(ext, B):
+@pytest.mark.parametrize("B,ksplit", [(20, 2), (7, 2), (20, 4), (7, 4)])
+def test_fused_bwd_matches_per_step_pair(ext, B, ksplit):
     """Fused hop+dgate backward == the per-step dgate/hop pair (same MFMA
     body, same f32 partial sum, same dgate math). B=7 exercises the odd
-    batch-row split (4 + 3 rows per pair block)."""
+    batch-row split; ksplit=4 the 4-way K-slice exchange (4 blocks per
+    n-tile, 5+5+5+5 / 2+2+2+1 row split)."""
     from zaremba_amd.models.lstm_lm import Model
     from zaremba_amd import trainer
 
     def grads(use_fused):
         torch.manual_seed(17)
         ext.set_use_fused_bwd(use_fused)
+        if use_fused:
+            ext.set_bwd_ksplit(ksplit)
         try:
             model = Model(60, 200, 2, dropout=0.0, winit=0.05,
                           engine="hip").to(dev())
@@ -241,14 +244,18 @@ def test_fused_bwd_matches_per_step_pair(ext, B):
             return {n: p.grad.clone() for n, p in model.named_parameters()}
         finally:
             ext.set_use_fused_bwd(True)
+            ext.set_bwd_ksplit(2)
 
     g1 = grads(True)
     g2 = grads(False)
     for n in g1:
         # not torch.equal: the bias/embedding grads go through atomicAdd
         # reductions whose summation order is nondeterministic run-to-run
-        # (~1e-13 wiggle) independent of the fused toggle
-        assert torch.allclose(g1[n], g2[n], atol=1e-8, rtol=1e-6), \
+        # (~1e-13 wiggle) independent of the fused toggle. The 4-way
+        # split's dh sum also reassociates the f32 partial adds.
+        tol = dict(atol=1e-8, rtol=1e-6) if ksplit == 2 else \
+            dict(atol=1e-6, rtol=1e-4)
+        assert torch.allclose(g1[n], g2[n], **tol), \
             (n, (g1[n] - g2[n]).abs().max().item())
 
 

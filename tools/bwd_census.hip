@@ -28,6 +28,7 @@ int main(int argc, char** argv) {
   int T = argc > 2 ? atoi(argv[2]) : 35;
   int B = argc > 3 ? atoi(argv[3]) : 20;
   int iters = argc > 4 ? atoi(argv[4]) : 20;
+  int nsplit = argc > 5 ? atoi(argv[5]) : 2;  // K-split ways (2 or 4)
   using zamd::bf16;
   const int HSp = 8;  // persistent_hs(H) for H >= 1500-ish; rec layout only
   const int K = 4 * H;
@@ -45,7 +46,7 @@ int main(int argc, char** argv) {
   CHK(hipMalloc(&dY, (size_t)T * B * H * 2));
   CHK(hipMalloc(&rec, rslot * T * 2));
   CHK(hipMalloc(&dG, (size_t)T * B * K * 2));
-  CHK(hipMalloc(&P, (size_t)2 * B * H * 4));
+  CHK(hipMalloc(&P, (size_t)4 * B * H * 4));
   CHK(hipMalloc(&dc, (size_t)B * H * 4));
   CHK(hipMalloc(&flags, (size_t)nbn * 4));
   CHK(hipMalloc(&abortf, 4));
@@ -70,7 +71,8 @@ int main(int argc, char** argv) {
       zamd::launch_smallm_fused_bwd(
           rd, W_pack, P, dY + (size_t)(t - 1) * B * H, dc,
           rec + (size_t)(t - 1) * rslot, dG + (size_t)(t - 1) * B * K, wr,
-          flags, abortf, B, H, K, HSp, (unsigned int)(T - t), nullptr);
+          flags, abortf, B, H, K, HSp, (unsigned int)(T - t), nsplit,
+          nullptr);
     }
   };
 
@@ -101,9 +103,9 @@ int main(int argc, char** argv) {
   const char* names[6] = {"hop-mfma",  "prefetch+red", "publish+drain",
                           "pair-sync", "dgate+stores", "entry-preamble"};
   const double launches = (double)iters * (T - 1);
-  const int grid = nbn * 2;
-  printf("H=%d T=%d B=%d grid=%d iters=%d abort=%u\n", H, T, B, grid, iters,
-         ab);
+  const int grid = nbn * nsplit;
+  printf("H=%d T=%d B=%d grid=%d ksplit=%d iters=%d abort=%u\n", H, T, B,
+         grid, nsplit, iters, ab);
   printf("wall: %.3f us/launch (%d launches/train)\n",
          wall_ms * 1000.0 / iters / (T - 1), T - 1);
   printf("%-14s %9s %9s %9s   (us/launch)\n", "phase", "mean", "min", "max");

@@ -152,11 +152,21 @@ static bool g_use_persistent = true;
 static bool g_use_persistent_bwd = false;  // see PERF.md: broadcast amplification loses
 
 static bool g_use_fused_bwd = true;
+// K-split ways for the fused backward hop (2 or 4). 4 runs ~2 blocks
+// per CU, interleaving the latency-bound A/W load chains (the hop-mfma
+// phase dominates the launch; census in PERF.md) — co-residency beyond
+// 1 block/CU comes from occupancy, and the bounded-spin abort flag
+// catches any scheduler that breaks it.
+static int g_bwd_ksplit = 2;
 
-// Fused backward step: every block pair must be co-resident (grid
-// 2*ceil(H/16) <= 256 CUs) and the batch fit the 32-row MFMA tile.
+// Fused backward step: every block of an n-tile's split group must be
+// co-resident (grid ksplit*ceil(H/16): guaranteed at <= 256 blocks = 1
+// per CU; relies on >= 2-blocks/CU occupancy above that) and the batch
+// must fit the 32-row MFMA tile.
 static bool fused_bwd_ok(int B, int H) {
-  return g_use_fused_bwd && B <= 32 && ((H + 15) / 16) * 2 <= 256;
+  const int nbn = (H + 15) / 16;
+  return g_use_fused_bwd && B <= 32 &&
+         nbn * g_bwd_ksplit <= (g_bwd_ksplit == 2 ? 256 : 400);
 }
 
 // The persistent forward needs every block co-resident and one cell
@@ -293,7 +303,7 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
       launch_smallm_fused_bwd(rd, WT_pack, dh_rec, dY + (t - 1) * hstep, dc,
                               rec + (t - 1) * rstep, dG + (t - 1) * gstep,
                               wr, flags, abort_flag, B, H, 4 * H, HSp,
-                              (unsigned int)(T - t), stream);
+                              (unsigned int)(T - t), g_bwd_ksplit, stream);
     }
     return;  // hop[0]'s dh output is unused (truncated-BPTT detach)
   }
@@ -336,6 +346,9 @@ static void lstm_seq_bwd(const torch::Tensor& dY, const torch::Tensor& gates,
                 "lstm_seq_bwd: hgran too small for fused-bwd counters");
     TORCH_CHECK(dG_pack.dim() == 2 && dG_pack.size(0) >= 2,
                 "lstm_seq_bwd: fused path needs >= 2 dG_pack slots");
+    TORCH_CHECK(dh_rec.numel() >= (int64_t)g_bwd_ksplit * B * H,
+                "lstm_seq_bwd: dh_rec too small for ", g_bwd_ksplit,
+                "-way K-split partials");
   }
   auto stream = current_stream();
   // Spin-synchronized kernels (fused pair / persistent) launch eagerly:
@@ -375,6 +388,10 @@ static void set_use_graphs(bool v) { g_use_graphs = v; }
 static void set_use_persistent(bool v) { g_use_persistent = v; }
 static void set_use_persistent_bwd(bool v) { g_use_persistent_bwd = v; }
 static void set_use_fused_bwd(bool v) { g_use_fused_bwd = v; }
+static void set_bwd_ksplit(int v) {
+  TORCH_CHECK(v == 2 || v == 4, "bwd ksplit must be 2 or 4");
+  g_bwd_ksplit = v;
+}
 static void clear_graphs() {
   for (auto& kv : g_fwd_graphs.cache) (void)hipGraphExecDestroy(kv.second);
   for (auto& kv : g_bwd_graphs.cache) (void)hipGraphExecDestroy(kv.second);
@@ -577,6 +594,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("persistent_hs", &zamd::persistent_hs);
   m.def("set_use_persistent_bwd", &zamd::set_use_persistent_bwd);
   m.def("set_use_fused_bwd", &zamd::set_use_fused_bwd);
+  m.def("set_bwd_ksplit", &zamd::set_bwd_ksplit);
   m.def("fused_bwd_active", &zamd::fused_bwd_ok);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

@@ -408,11 +408,13 @@ DEV_INLINE float load_wt_f32(const float* p) {
   return __builtin_bit_cast(float, u);
 }
 
-template <int MAXB>
+template <int MAXB, int NS>  // NS = K-split ways (2 or 4): 4 puts ~2
+                             // blocks on each CU, interleaving the
+                             // latency-bound A/W load chains
 __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     const bf16* __restrict__ A_pack,  // packed dG[t] (previous launch)
     const bf16* __restrict__ W_pack,  // packed W_h^T shadow
-    float* __restrict__ P,            // [2][M*N] f32 K-half partials
+    float* __restrict__ P,            // [NS][M*N] f32 K-slice partials
     const bf16* __restrict__ dy,      // dY[t-1], [B,H]
     float* __restrict__ dc,           // [B,H] carried cell grad
     const bf16* __restrict__ rec,     // rec[t-1] block records
@@ -430,12 +432,12 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   const int nbn = (N + 15) / 16;
   const int nb = blockIdx.x % nbn;
   const int n0 = nb * 16;
-  const int sk = blockIdx.x / nbn;  // K half (0/1)
+  const int sk = blockIdx.x / nbn;  // K slice (0..NS-1)
   const int w = wave_id();
   const int l = lane_id();
   const int lm = l & 15;
   const int KS = (K + 31) / 32;
-  const int KH = (KS + 1) / 2;
+  const int KH = (KS + NS - 1) / NS;
   const int ks0 = sk * KH;
   const int ks1 = min(ks0 + KH, KS);
 
@@ -478,7 +480,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   // dY, dc) NOW: the loads complete under the publish/arrive/spin that
   // follows instead of serializing into phase 2. nrows*16 <= 256, so
   // each thread owns at most one dgate element.
-  const int Bh = (M + 1) / 2;
+  const int Bh = (M + NS - 1) / NS;
   const int rb0 = sk * Bh;
   const int nrows = min(M - rb0, Bh);
   const int db = rb0 + threadIdx.x / 16;
@@ -529,7 +531,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     gau32* f = (gau32*)(uintptr_t)(flags + nb);
     __hip_atomic_fetch_add(f, 1u, ZRLX_AGENT);
     unsigned int spins = 0;
-    while (__hip_atomic_load(f, ZRLX_AGENT) < 2u * step) {
+    while (__hip_atomic_load(f, ZRLX_AGENT) < (unsigned int)NS * step) {
       __builtin_amdgcn_s_sleep(2);
       if (++spins > 20000000u) {
         atomicOr(abort_flag, 1u);
@@ -547,14 +549,19 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   // own 16 columns (mirrors lstm_cell_bwd_elt_kernel's math; inputs
   // preloaded above, only the partner partial is read here)
   if (dwork) {
-    const float* partner = P + (int64_t)(sk ^ 1) * M * N;
     const int b = db, j = dj, jj = dj - n0;
     const float vo = red[(0 * MAXB + b) * 16 + jj] +
                      red[(1 * MAXB + b) * 16 + jj] +
                      red[(2 * MAXB + b) * 16 + jj] +
                      red[(3 * MAXB + b) * 16 + jj];
     const int64_t e = (int64_t)b * N + j;
-    const float dh = p_dy + (vo + load_wt_f32(partner + e));
+    float ps = 0.f;  // partner partials; summed apart so the NS=2 total
+                     // stays p_dy + (vo + partner), bitwise-identical to
+                     // the per-step pair kernel's association
+#pragma unroll
+    for (int s = 1; s < NS; ++s)
+      ps += load_wt_f32(P + (int64_t)((sk + s) % NS) * M * N + e);
+    const float dh = p_dy + (vo + ps);
     const float do_ = dh * p_tc;
     const float dct = p_dc + dh * p_o * (1.f - p_tc * p_tc);
     const float di = dct * p_n;
@@ -586,11 +593,18 @@ void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
                              const bf16* rec, bf16* dG, bf16* dG_pack_out,
                              unsigned int* flags, unsigned int* abort_flag,
                              int M, int N, int K, int HSp, unsigned int step,
-                             hipStream_t stream) {
-  hipLaunchKernelGGL((smallm_fused_bwd_kernel<32>), dim3(cdiv(N, 16) * 2),
-                     dim3(CELL_THREADS), 0, stream, A_pack, W_pack, P, dy,
-                     dc, rec, dG, dG_pack_out, flags, abort_flag, M, N, K,
-                     HSp, step);
+                             int nsplit, hipStream_t stream) {
+  if (nsplit == 4) {
+    hipLaunchKernelGGL((smallm_fused_bwd_kernel<32, 4>),
+                       dim3(cdiv(N, 16) * 4), dim3(CELL_THREADS), 0, stream,
+                       A_pack, W_pack, P, dy, dc, rec, dG, dG_pack_out,
+                       flags, abort_flag, M, N, K, HSp, step);
+  } else {
+    hipLaunchKernelGGL((smallm_fused_bwd_kernel<32, 2>),
+                       dim3(cdiv(N, 16) * 2), dim3(CELL_THREADS), 0, stream,
+                       A_pack, W_pack, P, dy, dc, rec, dG, dG_pack_out,
+                       flags, abort_flag, M, N, K, HSp, step);
+  }
 }
 
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,

@@ -58,7 +58,9 @@ class _LayerWorkspace:
         self.gates = torch.zeros(T, B, 4 * H, dtype=bf, device=device)
         self.dY = torch.zeros(T, B, H, dtype=bf, device=device)
         self.dG = _slacked((T, B, 4 * H), bf, device, slack_ptrs)
-        self.dh_rec = torch.zeros(2, B, H, dtype=f32, device=device)
+        # K-split hop partials: up to 4 slices (ZAREMBA_AMD_BWD_KSPLIT);
+        # the per-step fallback pair uses slices 0/1
+        self.dh_rec = torch.zeros(4, B, H, dtype=f32, device=device)
         self.dc = torch.zeros(B, H, dtype=f32, device=device)
         # fragment-packed workspaces (zero-prefilled: pad rows/K-tails
         # must read as 0.0 in the packed cell kernels)
@@ -169,6 +171,8 @@ class HipModel:
             self.e.set_use_persistent(False)
         if os.environ.get("ZAREMBA_AMD_FUSED_BWD", "1") == "0":
             self.e.set_use_fused_bwd(False)
+        self.e.set_bwd_ksplit(
+            int(os.environ.get("ZAREMBA_AMD_BWD_KSPLIT", "2")))
         self.compute_dtype = torch.bfloat16
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,

@@ -42,7 +42,13 @@ def nll_loss(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
 
 def perplexity(data, model, batch_size: int) -> float:
     """Evaluate perplexity over a batched split (reference main.py:86-95):
-    fresh zero states, state carried across batches, exp(mean(loss/B))."""
+    fresh zero states, state carried across batches, exp(mean(loss/B)).
+
+    Under DP every rank evaluates the full split (replicated, not
+    sharded): the carried hidden state makes batch b's input state depend
+    on batch b-1, so sharding batches across ranks would change the
+    numbers vs the reference. Eval is a per-epoch cost (~1/13 of an
+    epoch's batches); replication keeps it semantics-exact."""
     was_training = model.training
     model.eval()
     with torch.no_grad():
@@ -185,6 +191,10 @@ def train(
                     "lr": lr, "elapsed_s": toc - tic,
                 })
         model.eval()
+        # Epoch boundary: surface any bounded-spin aborts before the eval
+        # result is trusted (log-step checks cover only cadence points).
+        if model._hip_model is not None:
+            model._hip_model.check_aborts()
         val_perp = perplexity(vld, model, batch_size)
         if is_rank0:
             print("Epoch : {:d} || Validation set perplexity : {:.3f}".format(
