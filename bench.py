@@ -149,7 +149,10 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if has_gpu else "fp32",
+            # HIP engine = bf16 compute / fp32 masters; the eager
+            # comparison path and CPU runs compute in fp32
+            "dtype": "bf16" if (has_gpu and args.engine != "eager")
+                     else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"zaremba-lstm-large-{args.layer_num}x{args.hidden_size}",
