@@ -29,6 +29,7 @@ int main(int argc, char** argv) {
   int B = argc > 3 ? atoi(argv[3]) : 20;
   int iters = argc > 4 ? atoi(argv[4]) : 20;
   int nsplit = argc > 5 ? atoi(argv[5]) : 2;  // K-split ways (2 or 4)
+  int nthr = argc > 6 ? atoi(argv[6]) * 64 : 256;  // waves/block (4 or 8)
   using zamd::bf16;
   const int HSp = 8;  // persistent_hs(H) for H >= 1500-ish; rec layout only
   const int K = 4 * H;
@@ -71,7 +72,7 @@ int main(int argc, char** argv) {
       zamd::launch_smallm_fused_bwd(
           rd, W_pack, P, dY + (size_t)(t - 1) * B * H, dc,
           rec + (size_t)(t - 1) * rslot, dG + (size_t)(t - 1) * B * K, wr,
-          flags, abortf, B, H, K, HSp, (unsigned int)(T - t), nsplit,
+          flags, abortf, B, H, K, HSp, (unsigned int)(T - t), nsplit, nthr,
           nullptr);
     }
   };
@@ -104,8 +105,8 @@ int main(int argc, char** argv) {
                           "pair-sync", "dgate+stores", "entry-preamble"};
   const double launches = (double)iters * (T - 1);
   const int grid = nbn * nsplit;
-  printf("H=%d T=%d B=%d grid=%d ksplit=%d iters=%d abort=%u\n", H, T, B,
-         grid, nsplit, iters, ab);
+  printf("H=%d T=%d B=%d grid=%d ksplit=%d thr=%d iters=%d abort=%u\n", H, T,
+         B, grid, nsplit, nthr, iters, ab);
   printf("wall: %.3f us/launch (%d launches/train)\n",
          wall_ms * 1000.0 / iters / (T - 1), T - 1);
   printf("%-14s %9s %9s %9s   (us/launch)\n", "phase", "mean", "min", "max");

@@ -158,6 +158,12 @@ static bool g_use_fused_bwd = true;
 // 1 block/CU comes from occupancy, and the bounded-spin abort flag
 // catches any scheduler that breaks it.
 static int g_bwd_ksplit = 2;
+// Threads per fused-bwd block (256 = 4 waves, 512 = 8): 8-wave blocks
+// double the in-flight load streams per CU at the same 1-block/CU grid
+// (the 4-way K-split A/B showed the hop is per-CU load-parallelism
+// bound but paid its gain back in sync skew + dispatch; census files
+// s2_census_ks*.txt).
+static int g_bwd_threads = 256;
 
 // Fused backward step: every block of an n-tile's split group must be
 // co-resident (grid ksplit*ceil(H/16): guaranteed at <= 256 blocks = 1
@@ -303,7 +309,8 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
       launch_smallm_fused_bwd(rd, WT_pack, dh_rec, dY + (t - 1) * hstep, dc,
                               rec + (t - 1) * rstep, dG + (t - 1) * gstep,
                               wr, flags, abort_flag, B, H, 4 * H, HSp,
-                              (unsigned int)(T - t), g_bwd_ksplit, stream);
+                              (unsigned int)(T - t), g_bwd_ksplit,
+                              g_bwd_threads, stream);
     }
     return;  // hop[0]'s dh output is unused (truncated-BPTT detach)
   }
@@ -391,6 +398,10 @@ static void set_use_fused_bwd(bool v) { g_use_fused_bwd = v; }
 static void set_bwd_ksplit(int v) {
   TORCH_CHECK(v == 2 || v == 4, "bwd ksplit must be 2 or 4");
   g_bwd_ksplit = v;
+}
+static void set_bwd_threads(int v) {
+  TORCH_CHECK(v == 256 || v == 512, "bwd threads must be 256 or 512");
+  g_bwd_threads = v;
 }
 static void clear_graphs() {
   for (auto& kv : g_fwd_graphs.cache) (void)hipGraphExecDestroy(kv.second);
@@ -595,6 +606,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_use_persistent_bwd", &zamd::set_use_persistent_bwd);
   m.def("set_use_fused_bwd", &zamd::set_use_fused_bwd);
   m.def("set_bwd_ksplit", &zamd::set_bwd_ksplit);
+  m.def("set_bwd_threads", &zamd::set_bwd_threads);
   m.def("fused_bwd_active", &zamd::fused_bwd_ok);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

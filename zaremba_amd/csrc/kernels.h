@@ -45,7 +45,7 @@ void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
                              const bf16* rec, bf16* dG, bf16* dG_pack_out,
                              unsigned int* flags, unsigned int* abort_flag,
                              int M, int N, int K, int HSp, unsigned int step,
-                             int nsplit, hipStream_t stream);
+                             int nsplit, int nthreads, hipStream_t stream);
 
 // lstm_persistent.hip — one launch for a whole layer unroll
 int persistent_hs(int H);

@@ -408,10 +408,14 @@ DEV_INLINE float load_wt_f32(const float* p) {
   return __builtin_bit_cast(float, u);
 }
 
-template <int MAXB, int NS>  // NS = K-split ways (2 or 4): 4 puts ~2
-                             // blocks on each CU, interleaving the
-                             // latency-bound A/W load chains
-__global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
+// NS = K-split ways (2 or 4): 4 puts ~2 blocks on each CU.
+// NTHR = threads/block: 512 (8 waves) doubles the in-flight load
+// streams per CU at the SAME grid — the measured hop bottleneck is
+// per-wave outstanding-load capacity (4-way K-split halved the hop
+// phase by doubling waves/CU, but paid for it in cross-block sync skew
+// and dispatch; 8-wave blocks get the load parallelism without either).
+template <int MAXB, int NS, int NTHR>
+__global__ __launch_bounds__(NTHR) void smallm_fused_bwd_kernel(
     const bf16* __restrict__ A_pack,  // packed dG[t] (previous launch)
     const bf16* __restrict__ W_pack,  // packed W_h^T shadow
     float* __restrict__ P,            // [NS][M*N] f32 K-slice partials
@@ -423,7 +427,8 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     unsigned int* __restrict__ flags, // [ceil(N/16)] monotonic counters
     unsigned int* __restrict__ abort_flag,
     int M, int N, int K, int HSp, unsigned int step) {
-  __shared__ float red[4 * MAXB * 16];
+  constexpr int NW = NTHR / 64;  // waves per block
+  __shared__ float red[NW * MAXB * 16];
   __shared__ int ok_s;
 
   BPROF_STAMP(bpe);  // first executed statement: kernel-entry stamp
@@ -447,13 +452,13 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
 
   BPROF_STAMP(bp0);
   f32x4 acc0 = {}, acc1 = {};
-  const int nown = (ks1 - ks0 - w + 3) / 4;
+  const int nown = (ks1 - ks0 - w + NW - 1) / NW;
   int i = 0;
   for (; i + 8 <= nown; i += 8) {
     bf16x8 a0v[8], a1v[8], bwv[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-      const int ks = ks0 + w + 4 * (i + u);
+      const int ks = ks0 + w + NW * (i + u);
       a0v[u] = pa[ks * 128];
       a1v[u] = pa[ks * 128 + 64];
       bwv[u] = pw[ks * 64];
@@ -465,7 +470,7 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
     }
   }
   for (; i < nown; ++i) {
-    const int ks = ks0 + w + 4 * i;
+    const int ks = ks0 + w + NW * i;
     bf16x8 a0v = pa[ks * 128];
     bf16x8 a1v = pa[ks * 128 + 64];
     bf16x8 bwv = pw[ks * 64];
@@ -515,11 +520,12 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   // publish this K-half's partial (write-through: memory-side visible
   // once the wave's vmcnt drains; the pair partner reads it sc1)
   float* mine = P + (int64_t)sk * M * N;
-  for (int idx = threadIdx.x; idx < M * 16; idx += CELL_THREADS) {
+  for (int idx = threadIdx.x; idx < M * 16; idx += NTHR) {
     const int b = idx / 16, jj = idx % 16;
     if (n0 + jj >= N) continue;
-    float v = red[(0 * MAXB + b) * 16 + jj] + red[(1 * MAXB + b) * 16 + jj] +
-              red[(2 * MAXB + b) * 16 + jj] + red[(3 * MAXB + b) * 16 + jj];
+    float v = 0.f;
+#pragma unroll
+    for (int ww = 0; ww < NW; ++ww) v += red[(ww * MAXB + b) * 16 + jj];
     store_wt_f32(mine + (int64_t)b * N + n0 + jj, v);
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -550,10 +556,9 @@ __global__ __launch_bounds__(CELL_THREADS) void smallm_fused_bwd_kernel(
   // preloaded above, only the partner partial is read here)
   if (dwork) {
     const int b = db, j = dj, jj = dj - n0;
-    const float vo = red[(0 * MAXB + b) * 16 + jj] +
-                     red[(1 * MAXB + b) * 16 + jj] +
-                     red[(2 * MAXB + b) * 16 + jj] +
-                     red[(3 * MAXB + b) * 16 + jj];
+    float vo = 0.f;
+#pragma unroll
+    for (int ww = 0; ww < NW; ++ww) vo += red[(ww * MAXB + b) * 16 + jj];
     const int64_t e = (int64_t)b * N + j;
     float ps = 0.f;  // partner partials; summed apart so the NS=2 total
                      // stays p_dy + (vo + partner), bitwise-identical to
@@ -593,18 +598,20 @@ void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
                              const bf16* rec, bf16* dG, bf16* dG_pack_out,
                              unsigned int* flags, unsigned int* abort_flag,
                              int M, int N, int K, int HSp, unsigned int step,
-                             int nsplit, hipStream_t stream) {
+                             int nsplit, int nthreads, hipStream_t stream) {
+#define ZAMD_FB_LAUNCH(NS_, NT_)                                           \
+  hipLaunchKernelGGL((smallm_fused_bwd_kernel<32, NS_, NT_>),              \
+                     dim3(cdiv(N, 16) * (NS_)), dim3(NT_), 0, stream,      \
+                     A_pack, W_pack, P, dy, dc, rec, dG, dG_pack_out,      \
+                     flags, abort_flag, M, N, K, HSp, step)
   if (nsplit == 4) {
-    hipLaunchKernelGGL((smallm_fused_bwd_kernel<32, 4>),
-                       dim3(cdiv(N, 16) * 4), dim3(CELL_THREADS), 0, stream,
-                       A_pack, W_pack, P, dy, dc, rec, dG, dG_pack_out,
-                       flags, abort_flag, M, N, K, HSp, step);
+    if (nthreads == 512) ZAMD_FB_LAUNCH(4, 512);
+    else                 ZAMD_FB_LAUNCH(4, 256);
   } else {
-    hipLaunchKernelGGL((smallm_fused_bwd_kernel<32, 2>),
-                       dim3(cdiv(N, 16) * 2), dim3(CELL_THREADS), 0, stream,
-                       A_pack, W_pack, P, dy, dc, rec, dG, dG_pack_out,
-                       flags, abort_flag, M, N, K, HSp, step);
+    if (nthreads == 512) ZAMD_FB_LAUNCH(2, 512);
+    else                 ZAMD_FB_LAUNCH(2, 256);
   }
+#undef ZAMD_FB_LAUNCH
 }
 
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
