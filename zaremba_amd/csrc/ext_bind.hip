@@ -159,11 +159,12 @@ static bool g_use_fused_bwd = true;
 // catches any scheduler that breaks it.
 static int g_bwd_ksplit = 2;
 // Threads per fused-bwd block (256 = 4 waves, 512 = 8): 8-wave blocks
-// double the in-flight load streams per CU at the same 1-block/CU grid
-// (the 4-way K-split A/B showed the hop is per-CU load-parallelism
-// bound but paid its gain back in sync skew + dispatch; census files
-// s2_census_ks*.txt).
-static int g_bwd_threads = 256;
+// double the in-flight load streams per CU at the same 1-block/CU grid.
+// Measured (profiles/s3_census_w*.txt): hop-mfma 6.32 -> 3.91 us with
+// sync/dispatch flat -> bench 269K -> 289K tokens/s (+7.3%); the 4-way
+// K-split alternative reached the same hop time but paid it all back in
+// cross-block sync skew + 376-block dispatch. Default 8 waves.
+static int g_bwd_threads = 512;
 
 // Fused backward step: every block of an n-tile's split group must be
 // co-resident (grid ksplit*ceil(H/16): guaranteed at <= 256 blocks = 1
