@@ -251,51 +251,66 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     PROF_STAMP(pt2);
     PROF_ACC(1, pt1, pt2);  // barrier (arrival drain + spin + acquire)
 #endif
-    // ---- stage h_t into the LDS image (plain loads, L2-amplified) --------
-    // 8-vector register chunks: all 8 loads issue before any ds_write
-    // consumes one. The previous load-then-write-per-vector loop paid
-    // one vmcnt wait per 16 B — 7.3 us/step-layer, HALF the kernel
-    // (tools/fwd_census). Loads are unguarded vec8: the last vector of
-    // a row over-reads <= 14 B into the next h_all row (in-bounds: the
+    // ---- stage h_t into LDS, software-pipelined against the MFMA ---------
+    // K is split at a granule boundary: half A is staged first (8/16-
+    // vector register chunks: all loads of a chunk issue before any
+    // ds_write consumes one — the old per-vector loop paid one vmcnt
+    // wait per 16 B, 7.3 us/step-layer); half B's sc1 loads are ISSUED
+    // into registers before the first sync, fly during the half-A MFMA
+    // (global vmcnt vs LDS lgkmcnt — no forced wait), and are written +
+    // consumed after it. Loads are unguarded vec8: the last vector of a
+    // row over-reads <= 14 B into the next h_all row (in-bounds: the
     // staging reads slots 0..T-1 of the T+1-slot buffer), and the hs
     // columns >= H it fills multiply Ws pad zeros in the MFMA.
+    const int vecs = (H + 7) / 8;
+    const int ksA = (KS + 1) / 2;  // granules in half A
+    const int vA = ksA * 4 < vecs ? ksA * 4 : vecs;  // vectors in half A
+    const int vB = vecs - vA;
+    const int totalB = B * vB;
+    bf16x8 vreg[16];
+    int nB = 0;  // this thread's half-B vector count
     {
       const bf16* hsrc = h_all + (int64_t)t * B * H;
-      const int vecs = (H + 7) / 8;
-      const int total = B * vecs;
-      for (int idx = t_; idx < total; idx += 16 * PCELL_THREADS) {
+      const int totalA = B * vA;
+      for (int idx = t_; idx < totalA; idx += 16 * PCELL_THREADS) {
         bf16x8 v[16];
         int id2 = idx;
 #pragma unroll
         for (int u = 0; u < 16; ++u, id2 += PCELL_THREADS) {
-          const int ic = id2 < total ? id2 : total - 1;
-          const int b = ic / vecs, k = (ic % vecs) * 8;
+          const int ic = id2 < totalA ? id2 : totalA - 1;
+          const int b = ic / vA, k = (ic % vA) * 8;
           v[u] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
         }
         id2 = idx;
 #pragma unroll
         for (int u = 0; u < 16; ++u, id2 += PCELL_THREADS) {
-          if (id2 < total) {
-            const int b = id2 / vecs, k = (id2 % vecs) * 8;
+          if (id2 < totalA) {
+            const int b = id2 / vA, k = (id2 % vA) * 8;
             *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = v[u];
           }
         }
       }
+      // issue half B's loads now; they complete under the half-A MFMA
+      for (int idx = t_; idx < totalB && nB < 16;
+           idx += PCELL_THREADS, ++nB) {
+        const int b = idx / vB, k = (vA + idx % vB) * 8;
+        vreg[nB] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
+      }
     }
-    __syncthreads();
+    __syncthreads();  // half A visible in LDS
 #ifdef ZAMD_FWD_PROF
     PROF_STAMP(pt3);
-    PROF_ACC(2, pt2, pt3);  // stage h into LDS
+    PROF_ACC(2, pt2, pt3);  // stage half A + issue half B
 #endif
 
-    // ---- gate MFMA reduction (wave g -> gate g) --------------------------
+    // ---- gate MFMA reduction (wave g -> gate g), half A ------------------
     f32x4 acc0 = {}, acc1 = {};
-    {
-      const bf16* pw = Ws + (int64_t)(g * HS + wc) * KPAD;
-      const bf16* pa0 = hs + (int64_t)a0r * KPAD;
-      const bf16* pa1 = hs + (int64_t)a1r * KPAD;
-      int ks = 0;
-      for (; ks + 4 <= KS; ks += 4) {
+    const bf16* pw = Ws + (int64_t)(g * HS + wc) * KPAD;
+    const bf16* pa0 = hs + (int64_t)a0r * KPAD;
+    const bf16* pa1 = hs + (int64_t)a1r * KPAD;
+    auto mfma_range = [&](int ks_lo, int ks_hi) {
+      int ks = ks_lo;
+      for (; ks + 4 <= ks_hi; ks += 4) {
         bf16x8 a0v[4], a1v[4], bwv[4];
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
@@ -310,7 +325,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
           acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
         }
       }
-      for (; ks < KS; ++ks) {
+      for (; ks < ks_hi; ++ks) {
         const int k = ks * 32 + lk;
         bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + k);
         bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + k);
@@ -318,6 +333,19 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
         acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
         acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
       }
+    };
+    mfma_range(0, ksA);
+    // ---- land half B in LDS, then finish the reduction -------------------
+    if (vB > 0) {
+      {
+        int idx = t_;
+        for (int u = 0; u < nB; ++u, idx += PCELL_THREADS) {
+          const int b = idx / vB, k = (vA + idx % vB) * 8;
+          *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = vreg[u];
+        }
+      }
+      __syncthreads();  // half B visible
+      mfma_range(ksA, KS);
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
