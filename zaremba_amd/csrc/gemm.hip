@@ -10,9 +10,12 @@
 //       register-starved and unused on the hot path).
 //
 // Design notes (measured on MI355X, see profiles/):
-//   * BMxBN output tile (128x128 or 64x64 chosen by grid size so skinny
-//     shapes still fill 256 CUs), BK=64, 4 waves in a 2x2 grid,
-//     v_mfma_f32_16x16x32_bf16 with fp32 accumulation.
+//   * TMxTN output tile with a WGMxWGN wave grid. Shapes pick their
+//     tile: 256x128 with 8 waves for the big dW/projection GEMMs
+//     (double arithmetic intensity + twice the staging streams),
+//     128x128 / 64x64 with 4 waves for mid/skinny shapes so the grid
+//     still fills the 256 CUs, 128x64 for the split-K dx family.
+//   * BK=64, v_mfma_f32_16x16x32_bf16 with fp32 accumulation.
 //   * Register-staged LDS with the T14 async-split schedule: issue tile
 //     t+1's global loads immediately after publishing tile t to LDS, so
 //     HBM/L2 latency hides under the MFMA phase.
@@ -21,14 +24,13 @@
 //     (one drained load per element), measured 3-5x slower.
 //   * LDS images are [row][k] with byte ^= (row&7)<<4 XOR swizzle:
 //     conflict-reduced ds_read_b128 column-slice reads.
-//   * The TN (transposed) staging loads 8 k-rows per thread with
+//   * The TN (transposed) staging loads k-rows per thread with
 //     lane-coalesced strided reads and writes one b128 per thread.
 #include "common.h"
 
 namespace zamd {
 
 constexpr int BK = 64;
-constexpr int GEMM_THREADS = 256;
 
 template <int BROWS>
 DEV_INLINE int swz(int row, int byte_col) {
@@ -47,22 +49,23 @@ DEV_INLINE void glds16(const bf16* gsrc, bf16* lds_dst_uniform) {
 }
 
 // Stage a [TILE][BK] k-contiguous tile via glds: TILE*BK*2/1024
-// instructions split across 4 waves. Out-of-range rows are CLAMPED to
-// the last valid row: the garbage values land only in output rows/cols
-// the epilogue discards (each output element depends only on its own
-// A row and B col), so M/N-edge blocks ride the same fast path — only
-// K-tail tiles need the guarded register staging (tail columns feed
+// instructions split across NWAVES waves. Out-of-range rows are CLAMPED
+// to the last valid row: the garbage values land only in output
+// rows/cols the epilogue discards (each output element depends only on
+// its own A row and B col), so M/N-edge blocks ride the same fast path —
+// only K-tail tiles need the guarded register staging (tail columns feed
 // REAL outputs and must be zero).
-template <int TILE>
+template <int TILE, int NWAVES>
 DEV_INLINE void stage_glds_kc(const bf16* __restrict__ src, int ld, int row0,
                               int nrows_total, int k0, bf16* lds) {
-  constexpr int NINST = TILE * BK * 2 / 1024;     // 16 (TILE=128) or 4
-  constexpr int PER_WAVE = NINST / 4;
+  constexpr int NINST = TILE * BK * 2 / 1024;  // 1 KB per instruction
+  constexpr int PER_WAVE = (NINST + NWAVES - 1) / NWAVES;
   const int w = wave_id();
   const int l = lane_id();
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     const int inst = w * PER_WAVE + i;
+    if (inst >= NINST) break;
     const int row = inst * 8 + (l >> 3);          // 8 rows per instruction
     const int gr = min(row0 + row, nrows_total - 1);
     const int colb = (l & 7) * 16;                // byte column 0..112
@@ -73,17 +76,19 @@ DEV_INLINE void stage_glds_kc(const bf16* __restrict__ src, int ld, int row0,
 }
 
 // ---- staging: k-contiguous operand (NT layout), [BROWS][BK] tile ---------
-// Register-load phase: each thread grabs BROWS*BK/(256*8) bf16x8 vectors.
-template <int BROWS, bool GUARD>
+// Register-load phase: each thread grabs BROWS*BK/(THREADS*8) bf16x8
+// vectors.
+template <int BROWS, int THREADS, bool GUARD>
 DEV_INLINE void stage_load_kc(const bf16* __restrict__ src, int ld, int row0,
                               int nrows, int k0, int K,
-                              bf16x8 (&v)[BROWS / 32]) {
+                              bf16x8 (&v)[BROWS * 8 / THREADS]) {
   const int t = threadIdx.x;
   const int kk = (t & 7) * 8;
-  const int r = t >> 3;
+  const int r = t >> 3;                 // 0 .. THREADS/8-1
+  constexpr int RSTRIDE = THREADS / 8;  // rows per pass
 #pragma unroll
-  for (int p = 0; p < BROWS / 32; ++p) {
-    const int row = r + p * 32;
+  for (int p = 0; p < BROWS * 8 / THREADS; ++p) {
+    const int row = r + p * RSTRIDE;
     if (GUARD) {
       // Branch-free guards: loads always execute from CLAMPED in-bounds
       // addresses; out-of-range values are zeroed by VALUE selects.
@@ -108,29 +113,30 @@ DEV_INLINE void stage_load_kc(const bf16* __restrict__ src, int ld, int row0,
   }
 }
 
-template <int BROWS>
-DEV_INLINE void stage_write_kc(bf16* lds, bf16x8 (&v)[BROWS / 32]) {
+template <int BROWS, int THREADS>
+DEV_INLINE void stage_write_kc(bf16* lds, bf16x8 (&v)[BROWS * 8 / THREADS]) {
   const int t = threadIdx.x;
   const int kk = (t & 7) * 8;
   const int r = t >> 3;
+  constexpr int RSTRIDE = THREADS / 8;
 #pragma unroll
-  for (int p = 0; p < BROWS / 32; ++p)
+  for (int p = 0; p < BROWS * 8 / THREADS; ++p)
     *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(lds) +
-                               swz<BROWS>(r + p * 32, kk * 2)) = v[p];
+                               swz<BROWS>(r + p * RSTRIDE, kk * 2)) = v[p];
 }
 
 // ---- staging: transposed operand (TN layout: src[K, rows]) ----------------
-template <int BROWS, bool GUARD>
+template <int BROWS, int THREADS, bool GUARD>
 DEV_INLINE void stage_load_tr(const bf16* __restrict__ src, int ld, int row0,
                               int nrows, int k0, int K,
-                              bf16x8 (&v)[BROWS / 32]) {
+                              bf16x8 (&v)[BROWS * 8 / THREADS]) {
   const int t = threadIdx.x;
   const int row = t & (BROWS - 1);
-  const int kb = (t / BROWS) * 8;           // 256/BROWS k-groups per pass
-  constexpr int KG = 8 * (256 / BROWS);     // k covered per pass
+  const int kb = (t / BROWS) * 8;           // THREADS/BROWS k-groups/pass
+  constexpr int KG = 8 * (THREADS / BROWS); // k covered per pass
   const int gr = row0 + row;
 #pragma unroll
-  for (int p = 0; p < BROWS / 32; ++p) {    // BK/KG passes == BROWS/32
+  for (int p = 0; p < BROWS * 8 / THREADS; ++p) {  // == BK/KG passes
     const int kk = kb + p * KG;
     if (GUARD) {
       // clamped-address loads + value selects (see stage_load_kc note)
@@ -149,14 +155,14 @@ DEV_INLINE void stage_load_tr(const bf16* __restrict__ src, int ld, int row0,
   }
 }
 
-template <int BROWS>
-DEV_INLINE void stage_write_tr(bf16* lds, bf16x8 (&v)[BROWS / 32]) {
+template <int BROWS, int THREADS>
+DEV_INLINE void stage_write_tr(bf16* lds, bf16x8 (&v)[BROWS * 8 / THREADS]) {
   const int t = threadIdx.x;
   const int row = t & (BROWS - 1);
   const int kb = (t / BROWS) * 8;
-  constexpr int KG = 8 * (256 / BROWS);
+  constexpr int KG = 8 * (THREADS / BROWS);
 #pragma unroll
-  for (int p = 0; p < BROWS / 32; ++p)
+  for (int p = 0; p < BROWS * 8 / THREADS; ++p)
     *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(lds) +
                                swz<BROWS>(row, (kb + p * KG) * 2)) = v[p];
 }
@@ -169,11 +175,12 @@ DEV_INLINE bf16x8 frag_read(const bf16* lds, int row, int k) {
 
 // ---------------------------------------------------------------------------
 // C[M,N] = A' @ B' + bias. TRANS_A: A is [K,M]; TRANS_B: B is [K,N];
-// otherwise the k-contiguous layouts A[M,K] / B[N,K]. BM=BN=TILE; 4 waves,
-// each (TILE/2)x(TILE/2).
+// otherwise the k-contiguous layouts A[M,K] / B[N,K]. TMxTN tile,
+// WGMxWGN wave grid (THREADS = WGM*WGN*64), each wave (TM/WGM)x(TN/WGN).
 // ---------------------------------------------------------------------------
-template <int TILE, bool TRANS_A, bool TRANS_B, typename OutT>
-__global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
+template <int TM, int TN, int WGM, int WGN, bool TRANS_A, bool TRANS_B,
+          typename OutT>
+__global__ __launch_bounds__(WGM* WGN * 64) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     OutT* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, OutT* __restrict__ C2, int kt_split) {
@@ -194,30 +201,35 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
       K = kt_split * BK;
     }
   }
-  constexpr int WT = TILE / 2;       // wave tile (64 or 32)
-  constexpr int NF = WT / 16;        // fragments per wave dim (4 or 2)
+  constexpr int THREADS = WGM * WGN * 64;
+  constexpr int WTM = TM / WGM;      // wave tile rows
+  constexpr int WTN = TN / WGN;      // wave tile cols
+  constexpr int NFM = WTM / 16;      // row fragments per wave
+  constexpr int NFN = WTN / 16;      // col fragments per wave
   // one LDS object only (a second __shared__ forces vmcnt(0) before every
-  // ds_read when a glds is in flight — guide §5 trap (a)); double-buffered
-  __shared__ bf16 smem_all[4 * TILE * BK];
-  bf16* As = smem_all;               // [2][TILE*BK] A buffers
-  bf16* Bs = smem_all + 2 * TILE * BK;
+  // ds_read when a glds is in flight — guide §5 trap (a)); double-buffered.
+  // Dynamic LDS: the 256x128 tile needs 96 KB, past the 64 KB static
+  // limit (gfx950 allows 160 KB/WG dynamically).
+  extern __shared__ __attribute__((aligned(16))) char gemm_smem[];
+  bf16* As = reinterpret_cast<bf16*>(gemm_smem);  // [2][TM*BK] A buffers
+  bf16* Bs = As + 2 * TM * BK;
 
-  const int nbn = (N + TILE - 1) / TILE;
+  const int nbn = (N + TN - 1) / TN;
   const int bm = blockIdx.x / nbn;
   const int bn = blockIdx.x % nbn;
-  const int m0 = bm * TILE, n0 = bn * TILE;
+  const int m0 = bm * TM, n0 = bn * TN;
   const bool interior =
-      (m0 + TILE <= M) && (n0 + TILE <= N);
+      (m0 + TM <= M) && (n0 + TN <= N);
 
   const int w = wave_id();
-  const int wm = (w >> 1) * WT;
-  const int wn = (w & 1) * WT;
+  const int wm = (w / WGN) * WTM;
+  const int wn = (w % WGN) * WTN;
   const int l = lane_id();
   const int lm = l & 15;
   const int lk = (l >> 4) * 8;
 
-  f32x4 acc[NF][NF] = {};
-  bf16x8 va[TILE / 32], vb[TILE / 32];
+  f32x4 acc[NFM][NFN] = {};
+  bf16x8 va[TM * 8 / THREADS], vb[TN * 8 / THREADS];
 
   const int nk = (K + BK - 1) / BK;
   const int k_full = K / BK;         // tiles with no K/MN guard
@@ -226,40 +238,44 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     const bool gk = (kt >= k_full);
     if (!guard_mn && !gk) {
       if (TRANS_A)
-        stage_load_tr<TILE, false>(A, lda, m0, TILE, kt * BK, K, va);
+        stage_load_tr<TM, THREADS, false>(A, lda, m0, TM, kt * BK, K, va);
       else
-        stage_load_kc<TILE, false>(A, lda, m0, TILE, kt * BK, K, va);
+        stage_load_kc<TM, THREADS, false>(A, lda, m0, TM, kt * BK, K, va);
       if (TRANS_B)
-        stage_load_tr<TILE, false>(B, ldb, n0, TILE, kt * BK, K, vb);
+        stage_load_tr<TN, THREADS, false>(B, ldb, n0, TN, kt * BK, K, vb);
       else
-        stage_load_kc<TILE, false>(B, ldb, n0, TILE, kt * BK, K, vb);
+        stage_load_kc<TN, THREADS, false>(B, ldb, n0, TN, kt * BK, K, vb);
     } else {
       if (TRANS_A)
-        stage_load_tr<TILE, true>(A, lda, m0, min(TILE, M - m0), kt * BK, K, va);
+        stage_load_tr<TM, THREADS, true>(A, lda, m0, min(TM, M - m0), kt * BK,
+                                         K, va);
       else
-        stage_load_kc<TILE, true>(A, lda, m0, min(TILE, M - m0), kt * BK, K, va);
+        stage_load_kc<TM, THREADS, true>(A, lda, m0, min(TM, M - m0), kt * BK,
+                                         K, va);
       if (TRANS_B)
-        stage_load_tr<TILE, true>(B, ldb, n0, min(TILE, N - n0), kt * BK, K, vb);
+        stage_load_tr<TN, THREADS, true>(B, ldb, n0, min(TN, N - n0), kt * BK,
+                                         K, vb);
       else
-        stage_load_kc<TILE, true>(B, ldb, n0, min(TILE, N - n0), kt * BK, K, vb);
+        stage_load_kc<TN, THREADS, true>(B, ldb, n0, min(TN, N - n0), kt * BK,
+                                         K, vb);
     }
   };
 
   auto mfma_phase = [&](const bf16* Ab, const bf16* Bb) {
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
-      bf16x8 af[NF], bfr[NF];
+      bf16x8 af[NFM], bfr[NFN];
 #pragma unroll
-      for (int i = 0; i < NF; ++i)
-        af[i] = frag_read<TILE>(Ab, wm + i * 16 + lm, kk + lk);
+      for (int i = 0; i < NFM; ++i)
+        af[i] = frag_read<TM>(Ab, wm + i * 16 + lm, kk + lk);
 #pragma unroll
-      for (int j = 0; j < NF; ++j)
-        bfr[j] = frag_read<TILE>(Bb, wn + j * 16 + lm, kk + lk);
+      for (int j = 0; j < NFN; ++j)
+        bfr[j] = frag_read<TN>(Bb, wn + j * 16 + lm, kk + lk);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int i = 0; i < NF; ++i)
+      for (int i = 0; i < NFM; ++i)
 #pragma unroll
-        for (int j = 0; j < NF; ++j)
+        for (int j = 0; j < NFN; ++j)
           acc[i][j] = mfma_16x16x32_bf16(af[i], bfr[j], acc[i][j]);
       __builtin_amdgcn_s_setprio(0);
     }
@@ -269,30 +285,31 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
     // glds 2-phase pipeline (guide §5.5 T3 minimum form): stage tile t+1
     // while computing tile t; __syncthreads() drains the in-flight glds.
     // Edge blocks use row-clamped sources (see stage_glds_kc).
-    stage_glds_kc<TILE>(A, lda, m0, M, 0, As);
-    stage_glds_kc<TILE>(B, ldb, n0, N, 0, Bs);
+    constexpr int NWAVES = WGM * WGN;
+    stage_glds_kc<TM, NWAVES>(A, lda, m0, M, 0, As);
+    stage_glds_kc<TN, NWAVES>(B, ldb, n0, N, 0, Bs);
     __syncthreads();
     int cur = 0;
     for (int kt = 0; kt < k_full; ++kt) {
       if (kt + 1 < k_full) {
-        stage_glds_kc<TILE>(A, lda, m0, M, (kt + 1) * BK,
-                            As + (cur ^ 1) * TILE * BK);
-        stage_glds_kc<TILE>(B, ldb, n0, N, (kt + 1) * BK,
-                            Bs + (cur ^ 1) * TILE * BK);
+        stage_glds_kc<TM, NWAVES>(A, lda, m0, M, (kt + 1) * BK,
+                                  As + (cur ^ 1) * TM * BK);
+        stage_glds_kc<TN, NWAVES>(B, ldb, n0, N, (kt + 1) * BK,
+                                  Bs + (cur ^ 1) * TN * BK);
       }
-      mfma_phase(As + cur * TILE * BK, Bs + cur * TILE * BK);
+      mfma_phase(As + cur * TM * BK, Bs + cur * TN * BK);
       __syncthreads();
       cur ^= 1;
     }
     if (k_full < nk) {  // K tail: register-staged, guarded
-      stage_load_kc<TILE, true>(A, lda, m0, min(TILE, M - m0), k_full * BK,
-                                K, va);
-      stage_load_kc<TILE, true>(B, ldb, n0, min(TILE, N - n0), k_full * BK,
-                                K, vb);
-      stage_write_kc<TILE>(As + cur * TILE * BK, va);
-      stage_write_kc<TILE>(Bs + cur * TILE * BK, vb);
+      stage_load_kc<TM, THREADS, true>(A, lda, m0, min(TM, M - m0),
+                                       k_full * BK, K, va);
+      stage_load_kc<TN, THREADS, true>(B, ldb, n0, min(TN, N - n0),
+                                       k_full * BK, K, vb);
+      stage_write_kc<TM, THREADS>(As + cur * TM * BK, va);
+      stage_write_kc<TN, THREADS>(Bs + cur * TN * BK, vb);
       __syncthreads();
-      mfma_phase(As + cur * TILE * BK, Bs + cur * TILE * BK);
+      mfma_phase(As + cur * TM * BK, Bs + cur * TN * BK);
     }
   } else {
     // register-staged path (edge blocks / transposed operands)
@@ -302,13 +319,13 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
       if (!ASYNC) load_tile(kt, !interior);
       __syncthreads();  // LDS consumers of tile kt-1 done
       if (TRANS_A)
-        stage_write_tr<TILE>(As, va);
+        stage_write_tr<TM, THREADS>(As, va);
       else
-        stage_write_kc<TILE>(As, va);
+        stage_write_kc<TM, THREADS>(As, va);
       if (TRANS_B)
-        stage_write_tr<TILE>(Bs, vb);
+        stage_write_tr<TN, THREADS>(Bs, vb);
       else
-        stage_write_kc<TILE>(Bs, vb);
+        stage_write_kc<TN, THREADS>(Bs, vb);
       if (ASYNC && kt + 1 < nk) load_tile(kt + 1, !interior);  // issue early
       __syncthreads();
       mfma_phase(As, Bs);
@@ -318,12 +335,12 @@ __global__ __launch_bounds__(GEMM_THREADS) void gemm_kernel(
   // Epilogue. C/D map: col = l&15, row = (l>>4)*4 + r.
   const int fr0 = (l >> 4) * 4;
 #pragma unroll
-  for (int j = 0; j < NF; ++j) {
+  for (int j = 0; j < NFN; ++j) {
     const int col = n0 + wn + j * 16 + (l & 15);
     if (col >= N) continue;
     const float bv = bias ? bias[col] : 0.f;
 #pragma unroll
-    for (int i = 0; i < NF; ++i) {
+    for (int i = 0; i < NFM; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm + i * 16 + fr0 + r;
@@ -337,34 +354,59 @@ template <bool TA, bool TB, typename OutT>
 void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
                    int M, int N, int K, int lda, int ldb, int ldc,
                    hipStream_t stream) {
-  // pick the tile: prefer 128^2; fall back to 64^2 when the grid would
-  // under-fill the 256 CUs (skinny backward shapes)
+  // pick the tile: prefer 256x128 (8 waves: 2x the arithmetic intensity
+  // AND 2x the staging streams of 128^2 — the dW/projection shapes were
+  // running at 0.3-0.5 PF on 128^2), then 128^2, falling back to 64^2
+  // when the grid would under-fill the 256 CUs (skinny backward shapes).
+  // The TN path keeps the 4-wave tiles (register-starved staging).
+  if (!TA && !TB) {
+    int grid256 = cdiv(M, 256) * cdiv(N, 128);
+    if (grid256 >= 192) {
+      hipLaunchKernelGGL((gemm_kernel<256, 128, 4, 2, TA, TB, OutT>),
+                         dim3(grid256), dim3(512),
+                         2 * (256 + 128) * BK * 2, stream, A, B, C, bias,
+                         M, N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
+      return;
+    }
+  }
   int grid128 = cdiv(M, 128) * cdiv(N, 128);
   if (grid128 >= 192) {
-    hipLaunchKernelGGL((gemm_kernel<128, TA, TB, OutT>), dim3(grid128),
-                       dim3(GEMM_THREADS), 0, stream, A, B, C, bias, M, N, K,
-                       lda, ldb, ldc, (OutT*)nullptr, 0);
+    hipLaunchKernelGGL((gemm_kernel<128, 128, 2, 2, TA, TB, OutT>),
+                       dim3(grid128), dim3(256), 2 * (128 + 128) * BK * 2,
+                       stream, A, B, C, bias, M,
+                       N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
   } else {
     int grid64 = cdiv(M, 64) * cdiv(N, 64);
-    hipLaunchKernelGGL((gemm_kernel<64, TA, TB, OutT>), dim3(grid64),
-                       dim3(GEMM_THREADS), 0, stream, A, B, C, bias, M, N, K,
-                       lda, ldb, ldc, (OutT*)nullptr, 0);
+    hipLaunchKernelGGL((gemm_kernel<64, 64, 2, 2, TA, TB, OutT>),
+                       dim3(grid64), dim3(256), 2 * (64 + 64) * BK * 2,
+                       stream, A, B, C, bias, M,
+                       N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
   }
 }
 
 // 2-way split-K NT GEMM writing partials C/C2 (caller combines). K must
 // be a BK multiple with >= 2 tiles (the k_pad contract guarantees it on
-// the dx shapes). 64-tile: the target shapes run ~1 block/CU, the z=2
-// grid puts 2 blocks/CU.
+// the dx shapes). 128x64 tile: the [700,1500] dx shapes give a 144-block
+// grid, x2 K-split = 288 (~1.1 blocks/CU) with half the B re-reads of
+// the old 64^2 (which ran 528 blocks re-pulling the W shadow per tile).
 template <typename OutT>
 void launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
                           const float* bias, int M, int N, int K, int lda,
                           int ldb, int ldc, hipStream_t stream) {
-  int grid64 = cdiv(M, 64) * cdiv(N, 64);
   int kt_split = (K / BK) / 2;
-  hipLaunchKernelGGL((gemm_kernel<64, false, false, OutT>),
-                     dim3(grid64, 1, 2), dim3(GEMM_THREADS), 0, stream, A, B,
-                     C, bias, M, N, K, lda, ldb, ldc, C2, kt_split);
+  int grid12864 = cdiv(M, 128) * cdiv(N, 64);
+  if (grid12864 * 2 >= 128) {
+    hipLaunchKernelGGL((gemm_kernel<128, 64, 2, 2, false, false, OutT>),
+                       dim3(grid12864, 1, 2), dim3(256),
+                       2 * (128 + 64) * BK * 2, stream, A, B, C,
+                       bias, M, N, K, lda, ldb, ldc, C2, kt_split);
+    return;
+  }
+  int grid64 = cdiv(M, 64) * cdiv(N, 64);
+  hipLaunchKernelGGL((gemm_kernel<64, 64, 2, 2, false, false, OutT>),
+                     dim3(grid64, 1, 2), dim3(256), 2 * (64 + 64) * BK * 2,
+                     stream, A, B, C, bias,
+                     M, N, K, lda, ldb, ldc, C2, kt_split);
 }
 
 #define INST(TA, TB, T)                                                     \

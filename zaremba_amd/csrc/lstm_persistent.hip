@@ -267,12 +267,13 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     const int vA = ksA * 4 < vecs ? ksA * 4 : vecs;  // vectors in half A
     const int vB = vecs - vA;
     const int totalB = B * vB;
-    // 8-vector chunks: each half is ~half the old 16-chunk payload, so 8
-    // keeps the same clamp-duplicate overhead per half (a fixed 16-deep
-    // chunk on the half measured stage-h 2.44 -> 4.48 us: ~8 DUPLICATE
-    // clamped sc1 loads per thread, each a real memory-side transaction)
+    // 8-vector chunks per half (each half is ~half the old 16-chunk
+    // payload, so the clamp-duplicate overhead stays the same). ALL
+    // register-array loops are compile-time unrolled with CLAMPED
+    // addresses: a runtime-bounded loop over a register array sends the
+    // array to scratch (measured: stage-h 2.4 -> 30 us, the whole
+    // kernel thrashing private memory).
     bf16x8 vreg[8];
-    int nB = 0;  // this thread's half-B vector count
     {
       const bf16* hsrc = h_all + (int64_t)t * B * H;
       const int totalA = B * vA;
@@ -295,10 +296,14 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
         }
       }
       // issue half B's loads now; they complete under the half-A MFMA
-      for (int idx = t_; idx < totalB && nB < 8;
-           idx += PCELL_THREADS, ++nB) {
-        const int b = idx / vB, k = (vA + idx % vB) * 8;
-        vreg[nB] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
+      if (vB > 0) {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int idx = t_ + u * PCELL_THREADS;
+          const int ic = idx < totalB ? idx : totalB - 1;
+          const int b = ic / vB, k = (vA + ic % vB) * 8;
+          vreg[u] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
+        }
       }
     }
     __syncthreads();  // half A visible in LDS
@@ -343,14 +348,18 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     if (vB > 0) {
       {
         const bf16* hsrc = h_all + (int64_t)t * B * H;
-        int idx = t_;
-        for (int u = 0; u < nB; ++u, idx += PCELL_THREADS) {
-          const int b = idx / vB, k = (vA + idx % vB) * 8;
-          *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = vreg[u];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int idx = t_ + u * PCELL_THREADS;
+          if (idx < totalB) {
+            const int b = idx / vB, k = (vA + idx % vB) * 8;
+            *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = vreg[u];
+          }
         }
         // overflow (totalB > 8 vectors/thread — only at B/H beyond the
         // Large config): stage the remainder load->write directly
-        for (; idx < totalB; idx += PCELL_THREADS) {
+        for (int idx = t_ + 8 * PCELL_THREADS; idx < totalB;
+             idx += PCELL_THREADS) {
           const int b = idx / vB, k = (vA + idx % vB) * 8;
           *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) =
               load_sc1_vec8(hsrc + (int64_t)b * H + k);
