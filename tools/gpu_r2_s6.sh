@@ -13,7 +13,8 @@ timeout 420 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/s6_prof -o
   python /root/repo/bench.py --gpus 1 --steps 30 --warmup 8 > /root/repo/gpurun_out/s6_prof_bench.json 2>/root/repo/gpurun_out/s6_prof.err
 echo "prof rc=$?" >> /root/repo/gpurun_out/s6_prof.err
 cd /root/repo
-python tools/prof_summary.py gpurun_out/s6_prof > gpurun_out/s6_kernel_stats.txt 2>&1 || true
+DB=$(ls gpurun_out/s6_prof/*results.db gpurun_out/s6_prof/**/*results.db 2>/dev/null | head -1)
+python tools/prof_summary.py "$DB" > gpurun_out/s6_kernel_stats.txt 2>&1 || true
 tail -n 12 gpurun_out/s6_fwd_census.txt
 cat gpurun_out/s6_bench.json
 head -n 24 gpurun_out/s6_kernel_stats.txt
