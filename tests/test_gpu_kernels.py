@@ -368,6 +368,29 @@ def test_lsm_nll_fwd_bwd_vs_eager(ext):
     assert (g1 - scores2.grad).abs().max().item() < 1e-5
 
 
+def test_embedding_bwd_atomic_vs_deterministic(ext):
+    """Race-detection A/B: the atomicAdd scatter-add path vs
+    the fixed-summation-order oracle. The oracle must be bitwise
+    reproducible across runs; the atomic path must match it numerically
+    (only fp32 add order differs)."""
+    torch.manual_seed(12)
+    V, H, N = 300, 1500, 700
+    dY = torch.randn(N, H, device=dev(), dtype=torch.bfloat16)
+    idx = torch.randint(0, V, (N,), device=dev())  # many duplicate rows
+    det1 = torch.zeros(V, H, device=dev())
+    det2 = torch.zeros(V, H, device=dev())
+    ato = torch.zeros(V, H, device=dev())
+    ext.embedding_bwd_det(dY, idx, det1)
+    ext.embedding_bwd_det(dY, idx, det2)
+    ext.embedding_bwd(dY, idx, ato)
+    assert torch.equal(det1, det2)  # deterministic: bitwise across runs
+    assert (ato - det1).abs().max().item() < 1e-4
+    # and both match the eager scatter-add reference
+    ref = torch.zeros(V, H, device=dev())
+    ref.index_add_(0, idx, dY.float())
+    assert (det1 - ref).abs().max().item() < 1e-4
+
+
 def test_softmax_acc_vs_eager(ext):
     """K13 fused ensemble accumulate: acc += softmax(scores) rowwise."""
     torch.manual_seed(11)

@@ -55,6 +55,37 @@ void launch_embedding_bwd(const bf16* dY, const int64_t* idx, float* dW,
                      dim3(waves_per_block * 64), 0, stream, dY, idx, dW, N, H);
 }
 
+// Deterministic embedding backward (the race-detection A/B oracle,
+// SURVEY.md §5): one wave per vocab row scans the index list IN ORDER
+// and accumulates matching dY rows — a fixed fp32 summation order, so
+// two runs are bitwise identical and the atomicAdd fast path can be
+// validated against it (numerically: the atomic path reorders the same
+// summands). O(V*N) index scans; validation/debug path, not the hot one.
+__global__ void embedding_bwd_det_kernel(const bf16* __restrict__ dY,
+                                         const int64_t* __restrict__ idx,
+                                         float* __restrict__ dW, int N, int H,
+                                         int V) {
+  int v = blockIdx.x * (blockDim.x / 64) + wave_id();
+  if (v >= V) return;
+  float* dst = dW + (int64_t)v * H;
+  int l = lane_id();
+  for (int n = 0; n < N; ++n) {
+    if ((int)idx[n] == v) {
+      const bf16* src = dY + (int64_t)n * H;
+      for (int k = l; k < H; k += 64) dst[k] += bf2f(src[k]);
+    }
+  }
+}
+
+void launch_embedding_bwd_det(const bf16* dY, const int64_t* idx, float* dW,
+                              int N, int H, int V, hipStream_t stream) {
+  int waves_per_block = 4;
+  int grid = cdiv(V, waves_per_block);
+  hipLaunchKernelGGL(embedding_bwd_det_kernel, dim3(grid),
+                     dim3(waves_per_block * 64), 0, stream, dY, idx, dW, N, H,
+                     V);
+}
+
 // ---------------------------------------------------------------------------
 // K5: dropout (inverted, scale 1/(1-p) at train time)
 // ---------------------------------------------------------------------------

@@ -482,6 +482,15 @@ static void embedding_bwd(const torch::Tensor& dY, const torch::Tensor& idx,
                        f_ptr_mut(dW), N, H, current_stream());
 }
 
+// deterministic (fixed summation order) variant: the race-detection A/B
+// oracle for the atomicAdd fast path
+static void embedding_bwd_det(const torch::Tensor& dY,
+                              const torch::Tensor& idx, torch::Tensor& dW) {
+  int N = idx.numel(), H = dW.size(1), V = dW.size(0);
+  launch_embedding_bwd_det(bf_ptr(dY), idx.data_ptr<int64_t>(),
+                           f_ptr_mut(dW), N, H, V, current_stream());
+}
+
 static void dropout_fwd(const torch::Tensor& x, torch::Tensor& y, double p,
                         int64_t seed, torch::Tensor& counter,
                         torch::Tensor& saved_offset) {
@@ -589,6 +598,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smallm_gemm_nt", &zamd::smallm_gemm_nt);
   m.def("embedding_fwd", &zamd::embedding_fwd);
   m.def("embedding_bwd", &zamd::embedding_bwd);
+  m.def("embedding_bwd_det", &zamd::embedding_bwd_det);
   m.def("dropout_fwd", &zamd::dropout_fwd);
   m.def("dropout_bwd", &zamd::dropout_bwd);
   m.def("lsm_nll_fwd", &zamd::lsm_nll_fwd);

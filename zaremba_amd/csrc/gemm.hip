@@ -206,6 +206,22 @@ __global__ __launch_bounds__(WGM* WGN * 64) void gemm_kernel(
   constexpr int WTN = TN / WGN;      // wave tile cols
   constexpr int NFM = WTM / 16;      // row fragments per wave
   constexpr int NFN = WTN / 16;      // col fragments per wave
+
+  // XCD-aware tile assignment: the dispatcher round-robins blockIdx
+  // across the 8 XCDs, so consecutive tiles (which share an A row-slice
+  // under the n-major tile order) land on DIFFERENT XCDs and every
+  // private L2 re-pulls nearly the whole operand set from LLC. Remap so
+  // each XCD owns a CONTIGUOUS tile stripe (a bijection handling the
+  // ragged tail: the first nblk%8 XCDs own ceil stripes).
+  int tile_id = blockIdx.x;
+  {
+    const int nblk = gridDim.x;
+    const int xcd = blockIdx.x & 7;
+    const int loc = blockIdx.x >> 3;
+    const int per_lo = nblk >> 3, r = nblk & 7;
+    tile_id = (xcd < r) ? xcd * (per_lo + 1) + loc
+                        : r * (per_lo + 1) + (xcd - r) * per_lo + loc;
+  }
   // one LDS object only (a second __shared__ forces vmcnt(0) before every
   // ds_read when a glds is in flight — guide §5 trap (a)); double-buffered.
   // Dynamic LDS: the 256x128 tile needs 96 KB, past the 64 KB static
@@ -215,8 +231,8 @@ __global__ __launch_bounds__(WGM* WGN * 64) void gemm_kernel(
   bf16* Bs = As + 2 * TM * BK;
 
   const int nbn = (N + TN - 1) / TN;
-  const int bm = blockIdx.x / nbn;
-  const int bn = blockIdx.x % nbn;
+  const int bm = tile_id / nbn;
+  const int bn = tile_id % nbn;
   const int m0 = bm * TM, n0 = bn * TN;
   const bool interior =
       (m0 + TM <= M) && (n0 + TN <= N);

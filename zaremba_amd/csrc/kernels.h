@@ -65,6 +65,8 @@ void launch_lstm_persistent_bwd(const bf16* dY, const bf16* rec,
 // elementwise.hip
 void launch_embedding_fwd(const bf16* W, const int64_t* idx, bf16* out,
                           int N, int H, hipStream_t stream);
+void launch_embedding_bwd_det(const bf16* dY, const int64_t* idx, float* dW,
+                              int N, int H, int V, hipStream_t stream);
 void launch_embedding_bwd(const bf16* dY, const int64_t* idx, float* dW,
                           int N, int H, hipStream_t stream);
 void launch_dropout_fwd(const bf16* x, bf16* y, float p, uint64_t seed,
