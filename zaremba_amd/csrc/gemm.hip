@@ -366,38 +366,47 @@ __global__ __launch_bounds__(WGM* WGN * 64) void gemm_kernel(
   }
 }
 
+// ZAMD_GEMM_TILE forces a tile for A/B sweeps: 256 (256x128/8w),
+// 128 (128^2/4w), 12864 (128x64/4w), 64 (64^2/4w); unset/0 = heuristic.
+static int force_tile() {
+  static int v = [] {
+    const char* e = getenv("ZAMD_GEMM_TILE");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
 template <bool TA, bool TB, typename OutT>
 void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
                    int M, int N, int K, int lda, int ldb, int ldc,
                    hipStream_t stream) {
-  // pick the tile: prefer 256x128 (8 waves: 2x the arithmetic intensity
-  // AND 2x the staging streams of 128^2 — the dW/projection shapes were
-  // running at 0.3-0.5 PF on 128^2), then 128^2, falling back to 64^2
-  // when the grid would under-fill the 256 CUs (skinny backward shapes).
-  // The TN path keeps the 4-wave tiles (register-starved staging).
+  // Tile choice is about CO-RESIDENT BLOCKS PER CU, not arithmetic
+  // intensity: probe data (profiles/s7_gemm_probe.txt) shows TF rising
+  // with grid size at constant shape family — these mid-size GEMMs are
+  // latency-bound. 256x128 (96 KB LDS -> 1 block/CU) measured no better
+  // than 128^2 (64 KB -> 2/CU); smaller tiles trade L2-dedup'd re-reads
+  // for more interleaved latency chains. The TN path keeps 4-wave tiles
+  // (register-starved staging).
+  const int f = force_tile();
+#define ZAMD_G_LAUNCH(TM_, TN_, WGM_, WGN_)                                   \
+  hipLaunchKernelGGL((gemm_kernel<TM_, TN_, WGM_, WGN_, TA, TB, OutT>),       \
+                     dim3(cdiv(M, TM_) * cdiv(N, TN_)),                       \
+                     dim3(WGM_ * WGN_ * 64), 2 * (TM_ + TN_) * BK * 2,        \
+                     stream, A, B, C, bias, M, N, K, lda, ldb, ldc,           \
+                     (OutT*)nullptr, 0)
   if (!TA && !TB) {
-    int grid256 = cdiv(M, 256) * cdiv(N, 128);
-    if (grid256 >= 192) {
-      hipLaunchKernelGGL((gemm_kernel<256, 128, 4, 2, TA, TB, OutT>),
-                         dim3(grid256), dim3(512),
-                         2 * (256 + 128) * BK * 2, stream, A, B, C, bias,
-                         M, N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
-      return;
-    }
+    if (f == 256) { ZAMD_G_LAUNCH(256, 128, 4, 2); return; }
+    if (f == 12864) { ZAMD_G_LAUNCH(128, 64, 2, 2); return; }
+    if (f == 128) { ZAMD_G_LAUNCH(128, 128, 2, 2); return; }
+    if (f == 64) { ZAMD_G_LAUNCH(64, 64, 2, 2); return; }
   }
   int grid128 = cdiv(M, 128) * cdiv(N, 128);
   if (grid128 >= 192) {
-    hipLaunchKernelGGL((gemm_kernel<128, 128, 2, 2, TA, TB, OutT>),
-                       dim3(grid128), dim3(256), 2 * (128 + 128) * BK * 2,
-                       stream, A, B, C, bias, M,
-                       N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
+    ZAMD_G_LAUNCH(128, 128, 2, 2);
   } else {
-    int grid64 = cdiv(M, 64) * cdiv(N, 64);
-    hipLaunchKernelGGL((gemm_kernel<64, 64, 2, 2, TA, TB, OutT>),
-                       dim3(grid64), dim3(256), 2 * (64 + 64) * BK * 2,
-                       stream, A, B, C, bias, M,
-                       N, K, lda, ldb, ldc, (OutT*)nullptr, 0);
+    ZAMD_G_LAUNCH(64, 64, 2, 2);
   }
+#undef ZAMD_G_LAUNCH
 }
 
 // 2-way split-K NT GEMM writing partials C/C2 (caller combines). K must
@@ -410,8 +419,12 @@ void launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
                           const float* bias, int M, int N, int K, int lda,
                           int ldb, int ldc, hipStream_t stream) {
   int kt_split = (K / BK) / 2;
-  int grid12864 = cdiv(M, 128) * cdiv(N, 64);
-  if (grid12864 * 2 >= 128) {
+  // 64^2 measured better in-train than 128x64 on the [700,1500] dx
+  // shapes (50.7 vs 54.8 us avg, s6 kernel stats): 528 tiny blocks
+  // interleave the latency chains harder than 288 mid ones.
+  // ZAMD_GEMM_TILE=12864 flips it for A/B.
+  if (force_tile() == 12864) {
+    int grid12864 = cdiv(M, 128) * cdiv(N, 64);
     hipLaunchKernelGGL((gemm_kernel<128, 64, 2, 2, false, false, OutT>),
                        dim3(grid12864, 1, 2), dim3(256),
                        2 * (128 + 64) * BK * 2, stream, A, B, C,
