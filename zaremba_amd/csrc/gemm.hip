@@ -399,6 +399,20 @@ void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
     if (f == 12864) { ZAMD_G_LAUNCH(128, 64, 2, 2); return; }
     if (f == 128) { ZAMD_G_LAUNCH(128, 128, 2, 2); return; }
     if (f == 64) { ZAMD_G_LAUNCH(64, 64, 2, 2); return; }
+    // Measured per-shape-family choices (profiles/s8_sweep.txt):
+    //  * tall-skinny dW gate shapes ([6000,1500,768]): 128x64 at ~4
+    //    blocks/CU, 32.3 vs 38.6 us on 128^2 (the >8192-M proj-dW
+    //    [10000,1500] flips back: 43.2 on 128^2 vs 45.5),
+    //  * wide-N forward shapes (proj [700,10000,1500] 46.9 vs 52.6;
+    //    input [700,6000,1500] 42.7 vs 45.8): 256x128/8-wave.
+    if (M >= 4 * N && M <= 8192 && cdiv(M, 128) * cdiv(N, 64) >= 384) {
+      ZAMD_G_LAUNCH(128, 64, 2, 2);
+      return;
+    }
+    if (N >= 2 * M && cdiv(M, 256) * cdiv(N, 128) >= 128) {
+      ZAMD_G_LAUNCH(256, 128, 4, 2);
+      return;
+    }
   }
   int grid128 = cdiv(M, 128) * cdiv(N, 128);
   if (grid128 >= 192) {
