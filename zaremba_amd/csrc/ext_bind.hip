@@ -401,7 +401,8 @@ static void set_bwd_ksplit(int v) {
   g_bwd_ksplit = v;
 }
 static void set_bwd_threads(int v) {
-  TORCH_CHECK(v == 256 || v == 512, "bwd threads must be 256 or 512");
+  TORCH_CHECK(v == 256 || v == 512 || v == 768,
+              "bwd threads must be 256, 512 or 768");
   g_bwd_threads = v;
 }
 static void clear_graphs() {
