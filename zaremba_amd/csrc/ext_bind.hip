@@ -164,7 +164,7 @@ static int g_bwd_ksplit = 2;
 // sync/dispatch flat -> bench 269K -> 289K tokens/s (+7.3%); the 4-way
 // K-split alternative reached the same hop time but paid it all back in
 // cross-block sync skew + 376-block dispatch. Default 8 waves.
-static int g_bwd_threads = 512;
+static int g_bwd_threads = 768;
 
 // Fused backward step: every block of an n-tile's split group must be
 // co-resident (grid ksplit*ceil(H/16): guaranteed at <= 256 blocks = 1
@@ -401,8 +401,8 @@ static void set_bwd_ksplit(int v) {
   g_bwd_ksplit = v;
 }
 static void set_bwd_threads(int v) {
-  TORCH_CHECK(v == 256 || v == 512 || v == 768,
-              "bwd threads must be 256, 512 or 768");
+  TORCH_CHECK(v == 256 || v == 512 || v == 768 || v == 1024,
+              "bwd threads must be 256, 512, 768 or 1024");
   g_bwd_threads = v;
 }
 static void clear_graphs() {

@@ -608,7 +608,8 @@ void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
     if (nthreads == 512) ZAMD_FB_LAUNCH(4, 512);
     else                 ZAMD_FB_LAUNCH(4, 256);
   } else {
-    if (nthreads == 768)      ZAMD_FB_LAUNCH(2, 768);
+    if (nthreads == 1024)     ZAMD_FB_LAUNCH(2, 1024);
+    else if (nthreads == 768) ZAMD_FB_LAUNCH(2, 768);
     else if (nthreads == 512) ZAMD_FB_LAUNCH(2, 512);
     else                      ZAMD_FB_LAUNCH(2, 256);
   }
