@@ -164,7 +164,7 @@ static int g_bwd_ksplit = 2;
 // sync/dispatch flat -> bench 269K -> 289K tokens/s (+7.3%); the 4-way
 // K-split alternative reached the same hop time but paid it all back in
 // cross-block sync skew + 376-block dispatch. Default 8 waves.
-static int g_bwd_threads = 768;
+static int g_bwd_threads = 1024;
 
 // Fused backward step: every block of an n-tile's split group must be
 // co-resident (grid ksplit*ceil(H/16): guaranteed at <= 256 blocks = 1

@@ -174,7 +174,7 @@ class HipModel:
         self.e.set_bwd_ksplit(
             int(os.environ.get("ZAREMBA_AMD_BWD_KSPLIT", "2")))
         self.e.set_bwd_threads(
-            int(os.environ.get("ZAREMBA_AMD_BWD_WAVES", "12")) * 64)
+            int(os.environ.get("ZAREMBA_AMD_BWD_WAVES", "16")) * 64)
         self.compute_dtype = torch.bfloat16
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,
