@@ -245,6 +245,9 @@ def test_fused_bwd_matches_per_step_pair(ext, B, ksplit, waves):
             s = model.state_init(B)
             scores, s = model(x, s)
             trainer.nll_loss(scores, y).backward()
+            # grads may still be in flight on the HipModel side stream
+            # (joined by sgd_step in real runs; tests read .grad directly)
+            torch.cuda.synchronize()
             return {n: p.grad.clone() for n, p in model.named_parameters()}
         finally:
             ext.set_use_fused_bwd(True)
@@ -283,6 +286,7 @@ def test_lstm_layer_autograd_matches_eager(ext):
     scores1, s1 = model(x, s1)
     loss1 = trainer.nll_loss(scores1, y)
     loss1.backward()
+    torch.cuda.synchronize()  # join the side-stream grad work
     import os
     os.environ["ZAREMBA_AMD_FORCE_EAGER"] = "1"
     try:

@@ -119,6 +119,8 @@ class _LayerRuntime:
         self.ws: Optional[_LayerWorkspace] = None
         self.slack_ptrs = None  # shared set, attached by HipModel
         self.dx_part = None  # f32 split-K partial pair, sized on first bwd
+        self.xT_buf = None   # persistent [Hin, TBp] transposed-x temp
+        self.owner = None    # back-ref to the HipModel (side stream)
 
     @torch.no_grad()
     def refresh(self, layer, e):
@@ -150,6 +152,7 @@ class _FcRuntime:
         self.dsc_buf = None  # slacked [N, V] bf16, sized on first backward
         self.dx_part = None  # f32 split-K partial pair, sized on first bwd
         self.slack_ptrs = None  # shared set, attached by HipModel
+        self.owner = None    # back-ref to the HipModel (side stream)
 
     @torch.no_grad()
     def refresh(self, fc, e):
@@ -192,6 +195,26 @@ class HipModel:
         self.dropout_counter = torch.zeros(1, dtype=torch.int64, device=dev)
         self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
         self._shadows_fresh = False
+        # Side stream for off-critical-path backward work (weight-grad
+        # transposes/GEMMs/colsums): they overlap the next layer's
+        # latency-bound fused backward train (which at 16 waves x 66
+        # VGPRs x 32 KB LDS leaves wave slots and LDS for concurrent
+        # GEMM blocks). The fused clip+SGD joins the recorded events
+        # before touching any grad. DISABLED under DP: the bucketer's
+        # post-accumulate hooks fire on the main stream and would race
+        # with side-stream grad production. NOTE: code reading .grad
+        # directly after backward() (instead of through sgd_step) must
+        # torch.cuda.synchronize() first.
+        self.side_stream = None
+        self.side_events: list = []
+        if os.environ.get("ZAREMBA_AMD_SIDE_STREAM", "1") == "1":
+            import torch.distributed as td
+            if not (td.is_available() and td.is_initialized()
+                    and td.get_world_size() > 1):
+                self.side_stream = torch.cuda.Stream(device=dev)
+        for rt in self.layers:
+            rt.owner = self
+        self.fc.owner = self
 
     def set_compute_dtype(self, dtype):
         if dtype != torch.bfloat16:
@@ -289,8 +312,19 @@ class HipModel:
         """
         m = self.model
         e = self.e
+        # join the side-stream weight-grad work before any grad is read
+        cur = torch.cuda.current_stream()
+        if self.side_events:
+            for ev in self.side_events:
+                cur.wait_event(ev)
+            self.side_events.clear()
         self.norm2.zero_()
         params = [p for p in m.parameters() if p.grad is not None]
+        if self.side_stream is not None:
+            # caching-allocator cross-stream rule: grads allocated on the
+            # side stream are consumed here on the main stream
+            for p in params:
+                p.grad.record_stream(cur)
         shadow_of = {id(m.embed.W): self.emb_W, id(m.fc.W): self.fc.W}
         for rt, layer in zip(self.layers, m.rnns):
             shadow_of[id(layer.W_x)] = rt.Wx

@@ -121,19 +121,39 @@ class LinearFn(torch.autograd.Function):
         # dW = dsc^T @ x via explicit transposes + the fast NT kernel (the
         # TN staging path is register-starved; measured 2-3x slower).
         # Transposed temps are zero-padded in K so the GEMM has no K-tail
-        # tile (a tail tile measured +13.5 us).
+        # tile (a tail tile measured +13.5 us). Off the critical path ->
+        # side stream (overlaps the LSTM layers' fused backward trains;
+        # see LstmLayerFn.backward).
         Np = ((N + 63) // 64) * 64
         if fc_rt.dscT is None or fc_rt.dscT.size(1) != Np:
             fc_rt.dscT = torch.zeros(V, Np, dtype=torch.bfloat16,
                                      device=x.device)
             fc_rt.xT = torch.zeros(H, Np, dtype=torch.bfloat16,
                                    device=x.device)
-        e.transpose_bf16(dsc, fc_rt.dscT)
-        e.transpose_bf16(x, fc_rt.xT)
-        dW = torch.empty(V, H, dtype=torch.float32, device=x.device)
-        e.gemm(fc_rt.dscT, fc_rt.xT, dW, None, False, False)
-        db = torch.zeros(V, dtype=torch.float32, device=x.device)
-        e.colsum_f32(dscores.contiguous(), db)
+        dscores_c = dscores.contiguous()
+
+        def dw_family():
+            e.transpose_bf16(dsc, fc_rt.dscT)
+            e.transpose_bf16(x, fc_rt.xT)
+            dW = torch.empty(V, H, dtype=torch.float32, device=x.device)
+            e.gemm(fc_rt.dscT, fc_rt.xT, dW, None, False, False)
+            db = torch.zeros(V, dtype=torch.float32, device=x.device)
+            e.colsum_f32(dscores_c, db)
+            return dW, db
+
+        hm = getattr(fc_rt, "owner", None)
+        side = hm.side_stream if hm is not None else None
+        if side is not None:
+            ev0 = torch.cuda.Event()
+            ev0.record()  # dsc / x / dscores ready on the main stream
+            side.wait_event(ev0)
+            with torch.cuda.stream(side):
+                dW, db = dw_family()
+                ev1 = torch.cuda.Event()
+                ev1.record()
+            hm.side_events.append(ev1)
+        else:
+            dW, db = dw_family()
         return dx, dW, db, None
 
 
@@ -222,21 +242,49 @@ class LstmLayerFn(torch.autograd.Function):
                        ws.abort)
         TB = T * B
         dG2 = ws.dG.view(TB, 4 * H)
+
         # dW_h = dG^T @ h_prev_stack ; dW_x = dG^T @ x ; dx = dG @ W_x —
         # weight grads via explicit transposes + the fast NT kernel.
         # K (=T*B) is zero-padded in the persistent transposed temps so
-        # no dW GEMM runs a K-tail tile.
-        dGT = ws.dGT
-        e.transpose_bf16(dG2, dGT)
-        hpT = ws.hpT
-        e.transpose_bf16(ws.h_all[:T].reshape(TB, H), hpT)
-        dWh = torch.empty(4 * H, H, dtype=torch.float32, device=x2.device)
-        e.gemm(dGT, hpT, dWh, None, False, False)
-        xT = torch.empty(Hin, ws.TBp, dtype=torch.bfloat16, device=x2.device)
-        xT[:, TB:].zero_()
-        e.transpose_bf16(x2.view(TB, Hin), xT)
-        dWx = torch.empty(4 * H, Hin, dtype=torch.float32, device=x2.device)
-        e.gemm(dGT, xT, dWx, None, False, False)
+        # no dW GEMM runs a K-tail tile. The whole dW family is OFF the
+        # critical path (consumed only by clip+SGD), so it runs on the
+        # HipModel side stream and overlaps the next layer's fused
+        # backward train; clip_and_sgd joins the events.
+        hm = getattr(rt, "owner", None)
+        side = hm.side_stream if hm is not None else None
+        if rt.xT_buf is None or rt.xT_buf.shape != (Hin, ws.TBp):
+            rt.xT_buf = torch.zeros(Hin, ws.TBp, dtype=torch.bfloat16,
+                                    device=x2.device)
+        xT = rt.xT_buf
+
+        def dw_family():
+            dGT = ws.dGT
+            e.transpose_bf16(dG2, dGT)
+            hpT = ws.hpT
+            e.transpose_bf16(ws.h_all[:T].reshape(TB, H), hpT)
+            dWh = torch.empty(4 * H, H, dtype=torch.float32,
+                              device=x2.device)
+            e.gemm(dGT, hpT, dWh, None, False, False)
+            e.transpose_bf16(x2.view(TB, Hin), xT)
+            dWx = torch.empty(4 * H, Hin, dtype=torch.float32,
+                              device=x2.device)
+            e.gemm(dGT, xT, dWx, None, False, False)
+            db = torch.zeros(4 * H, dtype=torch.float32, device=x2.device)
+            e.colsum_bf16(dG2, db)  # grads of b_x and b_h are identical
+            return dWh, dWx, db, db.clone()
+
+        if side is not None:
+            ev0 = torch.cuda.Event()
+            ev0.record()  # dG / h_all / x2 are ready on the main stream
+            side.wait_event(ev0)
+            with torch.cuda.stream(side):
+                dWh, dWx, db, db2 = dw_family()
+                ev1 = torch.cuda.Event()
+                ev1.record()
+            hm.side_events.append(ev1)
+        else:
+            dWh, dWx, db, db2 = dw_family()
+
         dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
         # dG is always slack-provisioned workspace; WxT carries zero pad
         # columns up to H4p from its strided transpose refresh. 2-way
@@ -253,7 +301,5 @@ class LstmLayerFn(torch.autograd.Function):
         else:
             e.gemm(dG2, rt.WxT, dx, None, False, False,
                    rt.H4p if rt.H4p != 4 * H else 0)
-        db = torch.zeros(4 * H, dtype=torch.float32, device=x2.device)
-        e.colsum_bf16(dG2, db)  # grads of b_x and b_h are identical
-        return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db.clone(),
+        return (dx.view(T, B, Hin), None, None, dWx, dWh, db, db2,
                 None)
