@@ -196,18 +196,20 @@ class HipModel:
         self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
         self._shadows_fresh = False
         # Side stream for off-critical-path backward work (weight-grad
-        # transposes/GEMMs/colsums): they overlap the next layer's
-        # latency-bound fused backward train (which at 16 waves x 66
-        # VGPRs x 32 KB LDS leaves wave slots and LDS for concurrent
-        # GEMM blocks). The fused clip+SGD joins the recorded events
-        # before touching any grad. DISABLED under DP: the bucketer's
-        # post-accumulate hooks fire on the main stream and would race
-        # with side-stream grad production. NOTE: code reading .grad
-        # directly after backward() (instead of through sgd_step) must
-        # torch.cuda.synchronize() first.
+        # transposes/GEMMs/colsums), overlapping the next layer's fused
+        # backward train; clip+SGD joins the recorded events. MEASURED
+        # AND REJECTED as the default (ZAREMBA_AMD_SIDE_STREAM=1 to
+        # re-enable): 302K vs 316K tokens/s OFF on the same box — the
+        # concurrent GEMM blocks contend with the co-resident spin
+        # kernels (slower pair partners extend every spin) and the
+        # event plumbing adds host latency to a launch-bound train.
+        # Also disabled under DP (the bucketer's post-accumulate hooks
+        # fire on the main stream and would race). NOTE when enabled:
+        # code reading .grad directly after backward() (instead of
+        # through sgd_step) must torch.cuda.synchronize() first.
         self.side_stream = None
         self.side_events: list = []
-        if os.environ.get("ZAREMBA_AMD_SIDE_STREAM", "1") == "1":
+        if os.environ.get("ZAREMBA_AMD_SIDE_STREAM", "0") == "1":
             import torch.distributed as td
             if not (td.is_available() and td.is_initialized()
                     and td.get_world_size() > 1):
