@@ -435,9 +435,14 @@ __global__ __launch_bounds__(NTHR) void smallm_fused_bwd_kernel(
   BPROF_LAUNCH_GAP(bpe);
 
   const int nbn = (N + 15) / 16;
-  const int nb = blockIdx.x % nbn;
+  // Partner-interleaved block order: the NS blocks of an n-tile are
+  // ADJACENT in blockIdx so the dispatcher starts them back-to-back.
+  // (With sk-major order partners launched ~nbn dispatch slots apart,
+  // and that start-time skew was the bulk of the 1.3 us pair-sync
+  // phase: the early partner spins until the late one publishes.)
+  const int nb = blockIdx.x / NS;
   const int n0 = nb * 16;
-  const int sk = blockIdx.x / nbn;  // K slice (0..NS-1)
+  const int sk = blockIdx.x % NS;   // K slice (0..NS-1)
   const int w = wave_id();
   const int l = lane_id();
   const int lm = l & 15;

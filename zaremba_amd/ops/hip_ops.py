@@ -58,7 +58,9 @@ class DropoutFn(torch.autograd.Function):
         # does not treat the reused buffer as an aliased input)
         x = x.contiguous()
         y = torch.empty_like(x) if out is None else out[0].view_as(x)
-        saved_offset = torch.zeros(1, dtype=torch.int64, device=x.device)
+        # empty, not zeros: dropout_tick overwrites it before any read
+        # (a zeros() here cost a 5 us fill launch x3 per step)
+        saved_offset = torch.empty(1, dtype=torch.int64, device=x.device)
         ext().dropout_fwd(x, y, p, seed, counter, saved_offset)
         ctx.p = p
         ctx.seed = seed
