@@ -74,12 +74,13 @@ def test_gemm_tn(ext, M, N, K):
 
 
 @pytest.mark.parametrize("M,N,K,k_pad", [
-    (700, 1500, 6000, 6016),   # lstm dx shape (k_pad path)
+    (700, 1500, 6000, 6016),   # lstm dx shape (k_pad path; 4-way)
     (700, 1500, 10048, 0),     # proj dx shape, K already a BK multiple
-    (130, 70, 256, 0),         # M/N edge blocks
+    (130, 70, 256, 0),         # M/N edge blocks (4 tiles -> 2-way)
+    (130, 70, 512, 0),         # 8 tiles -> 4-way with ragged last slice
 ])
 def test_gemm_splitk(ext, M, N, K, k_pad):
-    """2-way split-K partials + combine == plain NT GEMM result."""
+    """N-way split-K partials + combine == plain NT GEMM result."""
     torch.manual_seed(7)
     A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
     Keff = k_pad if k_pad else K
@@ -91,10 +92,11 @@ def test_gemm_splitk(ext, M, N, K, k_pad):
         flat[:M * K].copy_(A.reshape(-1))
         A = flat[:M * K].view(M, K)
     C1 = torch.empty(M, N, device=dev(), dtype=torch.float32)
-    C2 = torch.empty(M, N, device=dev(), dtype=torch.float32)
+    Cex = torch.empty(3, M, N, device=dev(), dtype=torch.float32)
     out = torch.empty(M, N, device=dev(), dtype=torch.bfloat16)
-    ext.gemm_splitk(A, B, C1, C2, None, k_pad)
-    ext.add2_f32_bf16(C1, C2, out)
+    nz = ext.gemm_splitk(A, B, C1, Cex, None, k_pad)
+    assert nz in (2, 4)
+    ext.addn_f32_bf16(C1, Cex, out, nz)
     ref = A.float() @ B[:, :K].float().t()
     assert rel_err(out, ref) < 5e-2, rel_err(out, ref)
 

@@ -56,13 +56,13 @@ if len(sys.argv) > 1 and sys.argv[1] == "splitk2":
         flat[:M*K].copy_(A.reshape(-1)); A = flat[:M*K].view(M, K)
         B = torch.zeros(N, Keff, device=dev, dtype=torch.bfloat16); B[:, :K].normal_()
         C1 = torch.empty(M, N, device=dev, dtype=torch.float32)
-        C2 = torch.empty(M, N, device=dev, dtype=torch.float32)
+        Cex = torch.empty(3, M, N, device=dev, dtype=torch.float32)
         out = torch.empty(M, N, device=dev, dtype=torch.bfloat16)
         for _ in range(5):
-            ext.gemm_splitk(A, B, C1, C2, None, k_pad); ext.add2_f32_bf16(C1, C2, out)
+            nz = ext.gemm_splitk(A, B, C1, Cex, None, k_pad); ext.addn_f32_bf16(C1, Cex, out, nz)
         torch.cuda.synchronize(); t0 = time.perf_counter()
         for _ in range(iters):
-            ext.gemm_splitk(A, B, C1, C2, None, k_pad); ext.add2_f32_bf16(C1, C2, out)
+            nz = ext.gemm_splitk(A, B, C1, Cex, None, k_pad); ext.addn_f32_bf16(C1, Cex, out, nz)
         torch.cuda.synchronize(); dt = (time.perf_counter()-t0)/iters
         print(f"SK M={M:6d} N={N:6d} K={K:6d} (pad {Keff}): {dt*1e6:8.1f} us  {2*M*N*K/dt/1e12:7.1f} TF")
     print("== split-K + combine vs plain ==")

@@ -604,4 +604,35 @@ void launch_add2_f32_bf16(const float* a, const float* b, bf16* out,
                      0, stream, a, b, out, n);
 }
 
+// combine a + the nextra stacked [n] f32 partial slices of `extra` into
+// bf16 (N-way split-K; exact MFMA accumulators, one rounding)
+__global__ void addn_f32_bf16_kernel(const float* __restrict__ a,
+                                     const float* __restrict__ extra,
+                                     int nextra, bf16* __restrict__ out,
+                                     int64_t n) {
+  int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i + 4 <= n) {
+    float4 v = *(const float4*)(a + i);
+    for (int p = 0; p < nextra; ++p) {
+      float4 e = *(const float4*)(extra + (int64_t)p * n + i);
+      v.x += e.x; v.y += e.y; v.z += e.z; v.w += e.w;
+    }
+    *(bf16x4*)(out + i) =
+        bf16x4{f2bf(v.x), f2bf(v.y), f2bf(v.z), f2bf(v.w)};
+  } else {
+    for (; i < n; ++i) {
+      float v = a[i];
+      for (int p = 0; p < nextra; ++p) v += extra[(int64_t)p * n + i];
+      out[i] = f2bf(v);
+    }
+  }
+}
+
+void launch_addn_f32_bf16(const float* a, const float* extra, int nextra,
+                          bf16* out, int64_t n, hipStream_t stream) {
+  int64_t thr = cdiv(n, 4);
+  hipLaunchKernelGGL(addn_f32_bf16_kernel, dim3(cdiv(thr, 256)), dim3(256),
+                     0, stream, a, extra, nextra, out, n);
+}
+
 }  // namespace zamd

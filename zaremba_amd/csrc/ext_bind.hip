@@ -103,9 +103,11 @@ static void gemm(const torch::Tensor& A, const torch::Tensor& B,
 // accumulators of the two K halves), combined by add2_f32_bf16. Same
 // k_pad contract as gemm(); the split keys off the padded K so both
 // halves are whole-BK-tile ranges.
-static void gemm_splitk(const torch::Tensor& A, const torch::Tensor& B,
-                        torch::Tensor& C, torch::Tensor& C2,
-                        const c10::optional<torch::Tensor>& bias, int k_pad) {
+// Returns the split ways nz used; C2 must hold >= nz-1 stacked [M, N]
+// partial slices (the caller sums C + C2[0..nz-2]).
+static int gemm_splitk(const torch::Tensor& A, const torch::Tensor& B,
+                       torch::Tensor& C, torch::Tensor& C2,
+                       const c10::optional<torch::Tensor>& bias, int k_pad) {
   int M = C.size(0), N = C.size(1);
   int K = A.size(1);
   TORCH_CHECK(A.size(0) == M, "gemm_splitk: A/C M mismatch");
@@ -113,15 +115,17 @@ static void gemm_splitk(const torch::Tensor& A, const torch::Tensor& B,
   int Keff = k_pad ? k_pad : K;
   TORCH_CHECK(B.size(1) == Keff, "gemm_splitk: B/K mismatch");
   TORCH_CHECK(Keff % 64 == 0 && Keff >= 128, "gemm_splitk: bad K");
-  TORCH_CHECK(C2.sizes() == C.sizes(), "gemm_splitk: C/C2 mismatch");
+  TORCH_CHECK(C2.numel() >= 3 * C.numel(),
+              "gemm_splitk: C2 must hold 3 stacked partial slices");
   TORCH_CHECK(C.scalar_type() == torch::kFloat32 &&
               C2.scalar_type() == torch::kFloat32,
               "gemm_splitk: f32 partials expected");
   const float* bp = bias ? f_ptr(*bias) : nullptr;
-  launch_gemm_splitk_t<float>(bf_ptr(A), bf_ptr(B), C.data_ptr<float>(),
-                              C2.data_ptr<float>(), bp, M, N, Keff,
-                              A.size(1), B.size(1), C.size(1),
-                              current_stream());
+  return launch_gemm_splitk_t<float>(bf_ptr(A), bf_ptr(B),
+                                     C.data_ptr<float>(),
+                                     C2.data_ptr<float>(), bp, M, N, Keff,
+                                     A.size(1), B.size(1), C.size(1),
+                                     current_stream());
 }
 
 static void add2_f32_bf16(const torch::Tensor& a, const torch::Tensor& b,
@@ -129,6 +133,15 @@ static void add2_f32_bf16(const torch::Tensor& a, const torch::Tensor& b,
   TORCH_CHECK(a.numel() == b.numel() && a.numel() == out.numel());
   launch_add2_f32_bf16(f_ptr(a), f_ptr(b), bf_ptr_mut(out), a.numel(),
                        current_stream());
+}
+
+// out = bf16(a + sum of the first (nparts-1) stacked slices of extra)
+static void addn_f32_bf16(const torch::Tensor& a, const torch::Tensor& extra,
+                          torch::Tensor& out, int nparts) {
+  TORCH_CHECK(nparts >= 2 && a.numel() == out.numel() &&
+              extra.numel() >= (nparts - 1) * a.numel());
+  launch_addn_f32_bf16(f_ptr(a), f_ptr(extra), nparts - 1, bf_ptr_mut(out),
+                       a.numel(), current_stream());
 }
 
 // ---------------------------------------------------------------------------
@@ -592,6 +605,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "2-way split-K NT GEMM (f32 partials)", py::arg("A"), py::arg("B"),
         py::arg("C"), py::arg("C2"), py::arg("bias"), py::arg("k_pad") = 0);
   m.def("add2_f32_bf16", &zamd::add2_f32_bf16);
+  m.def("addn_f32_bf16", &zamd::addn_f32_bf16);
   m.def("lstm_seq_fwd", &zamd::lstm_seq_fwd);
   m.def("lstm_seq_bwd", &zamd::lstm_seq_bwd);
   m.def("lstm_cell_fwd_step", &zamd::lstm_cell_fwd_step);

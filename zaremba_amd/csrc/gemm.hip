@@ -184,22 +184,22 @@ __global__ __launch_bounds__(WGM* WGN * 64) void gemm_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     OutT* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, OutT* __restrict__ C2, int kt_split) {
-  // 2-way split-K (kt_split > 0): blockIdx.z picks a K-tile range and a
-  // partial output (z0 -> C with bias, z1 -> C2 without). 2 co-resident
-  // blocks/CU interleave their latency chains on the ~1-block/CU dx
-  // shapes (measured: doubling the grid is near-free, so the split runs
-  // ~2x faster than one full-K pass). Caller combines C + C2.
+  // N-way split-K (kt_split > 0): blockIdx.z picks a K-tile range and a
+  // partial output (z0 -> C with bias; z>0 -> slice z-1 of the stacked
+  // C2 partials, no bias). gridDim.z co-resident blocks/CU interleave
+  // their latency chains on the ~1-block/CU dx shapes (measured:
+  // doubling the grid is near-free). Caller sums C + C2 slices.
   if (kt_split > 0) {
-    if (blockIdx.z) {
-      const int64_t koff = (int64_t)kt_split * BK;
+    const int z = blockIdx.z;
+    if (z) {
+      const int64_t koff = (int64_t)kt_split * BK * z;
       A += koff;
       B += koff;
-      K -= kt_split * BK;
-      C = C2;
+      K -= (int)(kt_split * BK * z);
+      C = C2 + (int64_t)(z - 1) * M * ldc;
       bias = nullptr;
-    } else {
-      K = kt_split * BK;
     }
+    if (z + 1 < (int)gridDim.z) K = kt_split * BK;
   }
   constexpr int THREADS = WGM * WGN * 64;
   constexpr int WTM = TM / WGM;      // wave tile rows
@@ -428,28 +428,32 @@ void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
 // the dx shapes). 128x64 tile: the [700,1500] dx shapes give a 144-block
 // grid, x2 K-split = 288 (~1.1 blocks/CU) with half the B re-reads of
 // the old 64^2 (which ran 528 blocks re-pulling the W shadow per tile).
+// Returns the split ways used (the caller sums that many partials).
 template <typename OutT>
-void launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
-                          const float* bias, int M, int N, int K, int lda,
-                          int ldb, int ldc, hipStream_t stream) {
-  int kt_split = (K / BK) / 2;
-  // 64^2 measured better in-train than 128x64 on the [700,1500] dx
-  // shapes (50.7 vs 54.8 us avg, s6 kernel stats): 528 tiny blocks
-  // interleave the latency chains harder than 288 mid ones.
-  // ZAMD_GEMM_TILE=12864 flips it for A/B.
+int launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
+                         const float* bias, int M, int N, int K, int lda,
+                         int ldb, int ldc, hipStream_t stream) {
+  // 4-way when K has >= 8 tiles (the dx shapes have 94-157): shorter
+  // per-block K chains + ~2 blocks/CU. 64^2 measured better in-train
+  // than 128x64 on the [700,1500] dx shapes (50.7 vs 54.8 us avg, s6
+  // kernel stats): tiny blocks interleave the latency chains harder.
+  // ZAMD_GEMM_TILE=12864 flips the tile for A/B.
+  const int nz = (K / BK) >= 8 ? 4 : 2;
+  int kt_split = (K / BK) / nz;
   if (force_tile() == 12864) {
     int grid12864 = cdiv(M, 128) * cdiv(N, 64);
     hipLaunchKernelGGL((gemm_kernel<128, 64, 2, 2, false, false, OutT>),
-                       dim3(grid12864, 1, 2), dim3(256),
+                       dim3(grid12864, 1, nz), dim3(256),
                        2 * (128 + 64) * BK * 2, stream, A, B, C,
                        bias, M, N, K, lda, ldb, ldc, C2, kt_split);
-    return;
+    return nz;
   }
   int grid64 = cdiv(M, 64) * cdiv(N, 64);
   hipLaunchKernelGGL((gemm_kernel<64, 64, 2, 2, false, false, OutT>),
-                     dim3(grid64, 1, 2), dim3(256), 2 * (64 + 64) * BK * 2,
+                     dim3(grid64, 1, nz), dim3(256), 2 * (64 + 64) * BK * 2,
                      stream, A, B, C, bias,
                      M, N, K, lda, ldb, ldc, C2, kt_split);
+  return nz;
 }
 
 #define INST(TA, TB, T)                                                     \
@@ -461,11 +465,11 @@ INST(false, false, bf16)
 INST(true, true, float)
 INST(true, true, bf16)
 #undef INST
-template void launch_gemm_splitk_t<float>(const bf16*, const bf16*, float*,
-                                          float*, const float*, int, int,
-                                          int, int, int, int, hipStream_t);
-template void launch_gemm_splitk_t<bf16>(const bf16*, const bf16*, bf16*,
-                                         bf16*, const float*, int, int, int,
-                                         int, int, int, hipStream_t);
+template int launch_gemm_splitk_t<float>(const bf16*, const bf16*, float*,
+                                         float*, const float*, int, int,
+                                         int, int, int, int, hipStream_t);
+template int launch_gemm_splitk_t<bf16>(const bf16*, const bf16*, bf16*,
+                                        bf16*, const float*, int, int, int,
+                                        int, int, int, hipStream_t);
 
 }  // namespace zamd

@@ -16,12 +16,13 @@ void launch_gemm_t(const bf16* A, const bf16* B, OutT* C, const float* bias,
                    int M, int N, int K, int lda, int ldb, int ldc,
                    hipStream_t stream);
 
-// 2-way split-K NT GEMM: one launch, grid.z=2, partials in C (with bias)
-// and C2 (without); K must be a BK(=64) multiple with >= 2 tiles.
+// N-way split-K NT GEMM: one launch, grid.z = nz (returned), partials in
+// C (with bias) and nz-1 stacked slices of C2 (without); K must be a
+// BK(=64) multiple with >= 2 tiles.
 template <typename OutT>
-void launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
-                          const float* bias, int M, int N, int K, int lda,
-                          int ldb, int ldc, hipStream_t stream);
+int launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
+                         const float* bias, int M, int N, int K, int lda,
+                         int ldb, int ldc, hipStream_t stream);
 
 // lstm.hip (fragment-packed operands; see lstm.hip header comment)
 void launch_pack_gated_w(const bf16* W, bf16* out, int rows, int ngates,
@@ -94,6 +95,8 @@ void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
                            int ldd, hipStream_t stream);
 void launch_add2_f32_bf16(const float* a, const float* b, bf16* out,
                           int64_t n, hipStream_t stream);
+void launch_addn_f32_bf16(const float* a, const float* extra, int nextra,
+                          bf16* out, int64_t n, hipStream_t stream);
 void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
                         hipStream_t stream);
 void launch_colsum_f32(const float* in, float* out, int R, int C,

@@ -106,17 +106,19 @@ class LinearFn(torch.autograd.Function):
         dsc = fc_rt.dsc_buf
         dsc.copy_(dscores)
         dx = torch.empty(N, H, dtype=torch.bfloat16, device=x.device)
-        # dx = dsc @ W: ~1 block/CU at 64-tile -> 2-way split-K (2 co-
-        # resident blocks/CU interleave the latency chains; measured
-        # near-2x), f32 partials combined with one rounding
+        # dx = dsc @ W: ~1 block/CU at 64-tile -> N-way split-K (co-
+        # resident blocks interleave the latency chains; measured
+        # near-free grid growth), f32 partials combined with one rounding
         if fc_rt.Vp >= 128:
             if fc_rt.dx_part is None or fc_rt.dx_part[0].shape != (N, H):
                 z = torch.empty(N, H, dtype=torch.float32, device=x.device)
-                fc_rt.dx_part = (z, torch.empty_like(z))
-            p1, p2 = fc_rt.dx_part
-            e.gemm_splitk(dsc, fc_rt.WT, p1, p2, None,
-                          fc_rt.Vp if fc_rt.Vp != V else 0)
-            e.add2_f32_bf16(p1, p2, dx)
+                fc_rt.dx_part = (z, torch.empty(3, N, H,
+                                                dtype=torch.float32,
+                                                device=x.device))
+            p1, pex = fc_rt.dx_part
+            nz = e.gemm_splitk(dsc, fc_rt.WT, p1, pex, None,
+                               fc_rt.Vp if fc_rt.Vp != V else 0)
+            e.addn_f32_bf16(p1, pex, dx, nz)
         else:
             e.gemm(dsc, fc_rt.WT, dx, None, False, False,
                    fc_rt.Vp if fc_rt.Vp != V else 0)
@@ -289,17 +291,19 @@ class LstmLayerFn(torch.autograd.Function):
 
         dx = torch.empty(TB, Hin, dtype=torch.bfloat16, device=x2.device)
         # dG is always slack-provisioned workspace; WxT carries zero pad
-        # columns up to H4p from its strided transpose refresh. 2-way
+        # columns up to H4p from its strided transpose refresh. N-way
         # split-K (see LinearFn.backward) on the ~1-block/CU shape.
         if rt.H4p >= 128:
             if rt.dx_part is None or rt.dx_part[0].shape != (TB, Hin):
                 z = torch.empty(TB, Hin, dtype=torch.float32,
                                 device=x2.device)
-                rt.dx_part = (z, torch.empty_like(z))
-            p1, p2 = rt.dx_part
-            e.gemm_splitk(dG2, rt.WxT, p1, p2, None,
-                          rt.H4p if rt.H4p != 4 * H else 0)
-            e.add2_f32_bf16(p1, p2, dx)
+                rt.dx_part = (z, torch.empty(3, TB, Hin,
+                                             dtype=torch.float32,
+                                             device=x2.device))
+            p1, pex = rt.dx_part
+            nz = e.gemm_splitk(dG2, rt.WxT, p1, pex, None,
+                               rt.H4p if rt.H4p != 4 * H else 0)
+            e.addn_f32_bf16(p1, pex, dx, nz)
         else:
             e.gemm(dG2, rt.WxT, dx, None, False, False,
                    rt.H4p if rt.H4p != 4 * H else 0)
