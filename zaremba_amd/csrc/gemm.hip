@@ -438,7 +438,11 @@ int launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
   // than 128x64 on the [700,1500] dx shapes (50.7 vs 54.8 us avg, s6
   // kernel stats): tiny blocks interleave the latency chains harder.
   // ZAMD_GEMM_TILE=12864 flips the tile for A/B.
-  const int nz = (K / BK) >= 8 ? 4 : 2;
+  static const int force_nz = [] {
+    const char* e = getenv("ZAMD_SPLITK_NZ");
+    return e ? atoi(e) : 0;
+  }();
+  const int nz = force_nz ? force_nz : ((K / BK) >= 8 ? 4 : 2);
   int kt_split = (K / BK) / nz;
   if (force_tile() == 12864) {
     int grid12864 = cdiv(M, 128) * cdiv(N, 64);

@@ -15,6 +15,7 @@ Owns, per model instance:
 
 from __future__ import annotations
 
+import os
 import secrets
 from typing import Optional
 
@@ -167,7 +168,6 @@ class HipModel:
         if self.device.type != "cuda":
             raise RuntimeError("HipModel requires a ROCm GPU device")
         self.e = _C.ext()
-        import os
         if os.environ.get("ZAREMBA_AMD_GRAPHS", "1") == "0":
             self.e.set_use_graphs(False)
         if os.environ.get("ZAREMBA_AMD_PERSISTENT", "1") == "0":
@@ -344,7 +344,8 @@ class HipModel:
             # norm chunks half-size: norm2_mt measured 4.2 TB/s at 1017
             # blocks (latency-bound per-block chains); 2x the blocks
             # shortens them. sgd_mt is already at the HBM roof.
-            CH, CHN = 65536, 32768
+            CH = 65536
+            CHN = int(os.environ.get("ZAREMBA_AMD_NORM_CHUNK", "32768"))
             nd, sd = [], []
             for p in params:
                 mp, gp = p.data.data_ptr(), p.grad.data_ptr()
