@@ -433,16 +433,17 @@ template <typename OutT>
 int launch_gemm_splitk_t(const bf16* A, const bf16* B, OutT* C, OutT* C2,
                          const float* bias, int M, int N, int K, int lda,
                          int ldb, int ldc, hipStream_t stream) {
-  // 4-way when K has >= 8 tiles (the dx shapes have 94-157): shorter
-  // per-block K chains + ~2 blocks/CU. 64^2 measured better in-train
-  // than 128x64 on the [700,1500] dx shapes (50.7 vs 54.8 us avg, s6
-  // kernel stats): tiny blocks interleave the latency chains harder.
-  // ZAMD_GEMM_TILE=12864 flips the tile for A/B.
+  // 2-way default: the same-box A/B matrix (profiles/s17_ab.txt)
+  // measured 4-way SLOWER end-to-end (321.7K vs 319.4K tokens/s) —
+  // the shorter K chains don't pay for the extra partial traffic +
+  // combine reads on these shapes. ZAMD_SPLITK_NZ=4 for A/B. 64^2
+  // tile (tiny blocks interleave latency chains harder than 128x64:
+  // 50.7 vs 54.8 us; ZAMD_GEMM_TILE=12864 flips it).
   static const int force_nz = [] {
     const char* e = getenv("ZAMD_SPLITK_NZ");
     return e ? atoi(e) : 0;
   }();
-  const int nz = force_nz ? force_nz : ((K / BK) >= 8 ? 4 : 2);
+  const int nz = force_nz ? force_nz : 2;
   int kt_split = (K / BK) / nz;
   if (force_tile() == 12864) {
     int grid12864 = cdiv(M, 128) * cdiv(N, 64);

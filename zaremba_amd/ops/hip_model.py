@@ -341,11 +341,12 @@ class HipModel:
             cache = self._mt_cache = {}
         descs = cache.get(key)
         if descs is None:
-            # norm chunks half-size: norm2_mt measured 4.2 TB/s at 1017
-            # blocks (latency-bound per-block chains); 2x the blocks
-            # shortens them. sgd_mt is already at the HBM roof.
+            # 64K chunks for BOTH: half-size norm chunks measured a
+            # 0.3-0.6% END-TO-END loss in the same-box A/B matrix
+            # (profiles/s17_ab.txt) despite the latency-bound-looking
+            # norm2 counters; ZAREMBA_AMD_NORM_CHUNK for A/B.
             CH = 65536
-            CHN = int(os.environ.get("ZAREMBA_AMD_NORM_CHUNK", "32768"))
+            CHN = int(os.environ.get("ZAREMBA_AMD_NORM_CHUNK", "65536"))
             nd, sd = [], []
             for p in params:
                 mp, gp = p.data.data_ptr(), p.grad.data_ptr()
