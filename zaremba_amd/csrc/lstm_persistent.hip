@@ -154,8 +154,11 @@ __device__ unsigned long long g_fwd_prof[256 * 8];
 // ===========================================================================
 // Forward
 // ===========================================================================
-template <int MAXB>
-__global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
+// NTHR: 256 = 4 waves; 512 = 8 waves where the EXTRA waves only help
+// the memory phases (h staging, publish, W preload) — the gate MFMA
+// stays on waves 0-3 so gbuf/LDS layout is unchanged.
+template <int MAXB, int NTHR>
+__global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
     const bf16* __restrict__ gx,     // [T, B, 4H]
     const bf16* __restrict__ W_h,    // [4H, H] bf16 shadow
     bf16* __restrict__ h_all,        // [T+1, B, H]; slot 0 = h0 (input)
@@ -182,7 +185,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
   const int t_ = threadIdx.x;
 
   // ---- load the block's W_h slice into LDS (once) ------------------------
-  for (int idx = t_; idx < 4 * HS * (KPAD / 8); idx += PCELL_THREADS) {
+  for (int idx = t_; idx < 4 * HS * (KPAD / 8); idx += NTHR) {
     const int kv = (idx % (KPAD / 8)) * 8;
     const int row = idx / (KPAD / 8);       // gg*HS + c
     const int gg = row / HS, c = row % HS;
@@ -199,7 +202,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     }
     *reinterpret_cast<bf16x8*>(Ws + (int64_t)row * KPAD + kv) = v;
   }
-  for (int idx = t_; idx < B * (KPAD - H); idx += PCELL_THREADS) {
+  for (int idx = t_; idx < B * (KPAD - H); idx += NTHR) {
     const int b = idx / (KPAD - H);
     const int k = H + idx % (KPAD - H);
     hs[(int64_t)b * KPAD + k] = (bf16)0.f;
@@ -273,22 +276,23 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     // addresses: a runtime-bounded loop over a register array sends the
     // array to scratch (measured: stage-h 2.4 -> 30 us, the whole
     // kernel thrashing private memory).
-    bf16x8 vreg[8];
+    constexpr int DEPTH = 2048 / NTHR;  // 8 at 256 thr, 4 at 512
+    bf16x8 vreg[DEPTH];
     {
       const bf16* hsrc = h_all + (int64_t)t * B * H;
       const int totalA = B * vA;
-      for (int idx = t_; idx < totalA; idx += 8 * PCELL_THREADS) {
-        bf16x8 v[8];
+      for (int idx = t_; idx < totalA; idx += DEPTH * NTHR) {
+        bf16x8 v[DEPTH];
         int id2 = idx;
 #pragma unroll
-        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+        for (int u = 0; u < DEPTH; ++u, id2 += NTHR) {
           const int ic = id2 < totalA ? id2 : totalA - 1;
           const int b = ic / vA, k = (ic % vA) * 8;
           v[u] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
         }
         id2 = idx;
 #pragma unroll
-        for (int u = 0; u < 8; ++u, id2 += PCELL_THREADS) {
+        for (int u = 0; u < DEPTH; ++u, id2 += NTHR) {
           if (id2 < totalA) {
             const int b = id2 / vA, k = (id2 % vA) * 8;
             *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = v[u];
@@ -298,8 +302,8 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
       // issue half B's loads now; they complete under the half-A MFMA
       if (vB > 0) {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int idx = t_ + u * PCELL_THREADS;
+        for (int u = 0; u < DEPTH; ++u) {
+          const int idx = t_ + u * NTHR;
           const int ic = idx < totalB ? idx : totalB - 1;
           const int b = ic / vB, k = (vA + ic % vB) * 8;
           vreg[u] = load_sc1_vec8(hsrc + (int64_t)b * H + k);
@@ -313,6 +317,8 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
 #endif
 
     // ---- gate MFMA reduction (wave g -> gate g), half A ------------------
+    // waves >= 4 (NTHR > 256) skip the MFMA: they exist for the memory
+    // phases; all barriers below are outside this guard
     f32x4 acc0 = {}, acc1 = {};
     const bf16* pw = Ws + (int64_t)(g * HS + wc) * KPAD;
     const bf16* pa0 = hs + (int64_t)a0r * KPAD;
@@ -343,35 +349,38 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
         acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
       }
     };
-    mfma_range(0, ksA);
+    if (g < 4) mfma_range(0, ksA);
     // ---- land half B in LDS, then finish the reduction -------------------
     if (vB > 0) {
       {
         const bf16* hsrc = h_all + (int64_t)t * B * H;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int idx = t_ + u * PCELL_THREADS;
+        for (int u = 0; u < DEPTH; ++u) {
+          const int idx = t_ + u * NTHR;
           if (idx < totalB) {
             const int b = idx / vB, k = (vA + idx % vB) * 8;
             *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) = vreg[u];
           }
         }
-        // overflow (totalB > 8 vectors/thread — only at B/H beyond the
-        // Large config): stage the remainder load->write directly
-        for (int idx = t_ + 8 * PCELL_THREADS; idx < totalB;
-             idx += PCELL_THREADS) {
+        // overflow (totalB > DEPTH vectors/thread — only at B/H beyond
+        // the Large config): stage the remainder load->write directly
+        for (int idx = t_ + DEPTH * NTHR; idx < totalB;
+             idx += NTHR) {
           const int b = idx / vB, k = (vA + idx % vB) * 8;
           *reinterpret_cast<bf16x8*>(hs + (int64_t)b * KPAD + k) =
               load_sc1_vec8(hsrc + (int64_t)b * H + k);
         }
       }
       __syncthreads();  // half B visible
-      mfma_range(ksA, KS);
+      if (g < 4) mfma_range(ksA, KS);
     }
+    if (g < 4) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      if (fr0 + r < B) gbuf[(g * B + fr0 + r) * 16 + lm] = acc0[r];
-      if (16 + fr0 + r < B) gbuf[(g * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
+      for (int r = 0; r < 4; ++r) {
+        if (fr0 + r < B) gbuf[(g * B + fr0 + r) * 16 + lm] = acc0[r];
+        if (16 + fr0 + r < B)
+          gbuf[(g * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
+      }
     }
     __syncthreads();
 #ifdef ZAMD_FWD_PROF
@@ -415,7 +424,7 @@ __global__ __launch_bounds__(PCELL_THREADS) void lstm_persistent_fwd_kernel(
     // ---- publish h_{t+1}: paired write-through stores --------------------
     {
       bf16* hdst = h_all + (int64_t)(t + 1) * B * H;
-      for (int i = t_; i < B * HS / 2; i += PCELL_THREADS) {
+      for (int i = t_; i < B * HS / 2; i += NTHR) {
         const int b = i / (HS / 2);
         const int jj = (i % (HS / 2)) * 2;
         const int j = j0 + jj;
@@ -439,12 +448,22 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
                                 int H, hipStream_t stream) {
   const int HS = persistent_hs(H);
   const int NB = cdiv(H, HS);
-  const int KS = (H + 31) / 32;
-  const int KPAD = KS * 32 + 8;
   size_t lds = persistent_fwd_lds(B, H);
-  hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32>), dim3(NB),
-                     dim3(PCELL_THREADS), lds, stream, gx, W_h, h_all, c_all,
-                     rec, pstate, abort_flag, T, B, H, HS);
+  // ZAREMBA_AMD_FWD_WAVES: 4 (default) or 8 — the 8-wave variant gives
+  // the memory phases twice the streams at unchanged MFMA/LDS layout.
+  static const int nthr = [] {
+    const char* e = getenv("ZAREMBA_AMD_FWD_WAVES");
+    return e ? atoi(e) * 64 : 256;
+  }();
+  if (nthr == 512) {
+    hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32, 512>), dim3(NB),
+                       dim3(512), lds, stream, gx, W_h, h_all, c_all,
+                       rec, pstate, abort_flag, T, B, H, HS);
+  } else {
+    hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32, 256>), dim3(NB),
+                       dim3(256), lds, stream, gx, W_h, h_all, c_all,
+                       rec, pstate, abort_flag, T, B, H, HS);
+  }
 }
 
 // ===========================================================================
