@@ -449,11 +449,12 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
   const int HS = persistent_hs(H);
   const int NB = cdiv(H, HS);
   size_t lds = persistent_fwd_lds(B, H);
-  // ZAREMBA_AMD_FWD_WAVES: 4 (default) or 8 — the 8-wave variant gives
-  // the memory phases twice the streams at unchanged MFMA/LDS layout.
+  // ZAREMBA_AMD_FWD_WAVES: 8 (default) or 4 — 8-wave blocks give the
+  // memory phases twice the streams at unchanged MFMA/LDS layout
+  // (measured 329.0K vs 324.7K tokens/s same-box; s18 censuses).
   static const int nthr = [] {
     const char* e = getenv("ZAREMBA_AMD_FWD_WAVES");
-    return e ? atoi(e) * 64 : 256;
+    return e ? atoi(e) * 64 : 512;
   }();
   if (nthr == 512) {
     hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32, 512>), dim3(NB),
