@@ -194,15 +194,40 @@ __global__ void lsm_nll_fwd_kernel(const float* __restrict__ scores,
   int row = blockIdx.x;
   if (row >= N) return;
   const float* s = scores + (int64_t)row * V;
+  // Two independent online chains per thread (contiguous halves): the
+  // single-chain loop was a serial dependent update per strided load —
+  // latency-bound at ~1.2 TB/s over the [700,10000] scores row.
+  const int half = (V >= 2 * (int)blockDim.x)
+                       ? (int)(((V + 2 * blockDim.x - 1) / (2 * blockDim.x)) *
+                               blockDim.x)
+                       : V;
   float m = -INFINITY, acc = 0.f;
-  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+  float m1 = -INFINITY, a1 = 0.f;
+  for (int v = threadIdx.x; v < half; v += blockDim.x) {
     float x = s[v];
+    const int v2 = v + half;
+    float x1 = s[min(v2, V - 1)];  // clamped load; guarded update below
     if (x > m) {
       acc = acc * __expf(m - x) + 1.f;
       m = x;
     } else {
       acc += __expf(x - m);
     }
+    if (v2 < V) {
+      if (x1 > m1) {
+        a1 = a1 * __expf(m1 - x1) + 1.f;
+        m1 = x1;
+      } else {
+        a1 += __expf(x1 - m1);
+      }
+    }
+  }
+  {  // merge the two chains (guard the never-ran -inf case)
+    float mn = fmaxf(m, m1);
+    float pa = (m == -INFINITY) ? 0.f : acc * __expf(m - mn);
+    float pb = (m1 == -INFINITY) ? 0.f : a1 * __expf(m1 - mn);
+    acc = pa + pb;
+    m = mn;
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
@@ -540,15 +565,25 @@ void launch_transpose_bf16(const bf16* src, bf16* dst, int R, int C,
 // column sum of a bf16 matrix -> f32 (bias grads: db = colsum(dG)).
 // 2-D grid: row-chunks in y accumulate via atomicAdd (out pre-zeroed by
 // the wrapper); a single serial column walk was latency-bound (165 us).
+// 4 independent accumulators: the single-acc loop was a serial
+// dependent-add chain over C-strided (cache-line-apart) loads —
+// 1.65 TB/s; four chains overlap the load latencies.
 __global__ void colsum_bf16_kernel(const bf16* __restrict__ in,
                                    float* __restrict__ out, int Rr, int Cc) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= Cc) return;
   int r0 = blockIdx.y * 64;
   int r1 = min(r0 + 64, Rr);
-  float acc = 0.f;
-  for (int r = r0; r < r1; ++r) acc += bf2f(in[(int64_t)r * Cc + c]);
-  atomicAdd(out + c, acc);
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int r = r0;
+  for (; r + 4 <= r1; r += 4) {
+    a0 += bf2f(in[(int64_t)(r + 0) * Cc + c]);
+    a1 += bf2f(in[(int64_t)(r + 1) * Cc + c]);
+    a2 += bf2f(in[(int64_t)(r + 2) * Cc + c]);
+    a3 += bf2f(in[(int64_t)(r + 3) * Cc + c]);
+  }
+  for (; r < r1; ++r) a0 += bf2f(in[(int64_t)r * Cc + c]);
+  atomicAdd(out + c, (a0 + a1) + (a2 + a3));
 }
 
 void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
@@ -559,16 +594,23 @@ void launch_colsum_bf16(const bf16* in, float* out, int R, int C,
 }
 
 // f32 column sum (projection db = colsum(dscores), f32 [N,V] input):
-// same row-chunked atomicAdd scheme as the bf16 variant.
+// same row-chunked atomicAdd scheme + 4-chain ILP as the bf16 variant.
 __global__ void colsum_f32_kernel(const float* __restrict__ in,
                                   float* __restrict__ out, int Rr, int Cc) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= Cc) return;
   int r0 = blockIdx.y * 64;
   int r1 = min(r0 + 64, Rr);
-  float acc = 0.f;
-  for (int r = r0; r < r1; ++r) acc += in[(int64_t)r * Cc + c];
-  atomicAdd(out + c, acc);
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int r = r0;
+  for (; r + 4 <= r1; r += 4) {
+    a0 += in[(int64_t)(r + 0) * Cc + c];
+    a1 += in[(int64_t)(r + 1) * Cc + c];
+    a2 += in[(int64_t)(r + 2) * Cc + c];
+    a3 += in[(int64_t)(r + 3) * Cc + c];
+  }
+  for (; r < r1; ++r) a0 += in[(int64_t)r * Cc + c];
+  atomicAdd(out + c, (a0 + a1) + (a2 + a3));
 }
 
 void launch_colsum_f32(const float* in, float* out, int R, int C,
