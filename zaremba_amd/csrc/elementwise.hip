@@ -537,20 +537,37 @@ __global__ void transpose_bf16_kernel(const bf16* __restrict__ src,
     }
   }
   __syncthreads();
-  // write: thread covers (col, 8 rows): dst row = c0+col, cols r0+r8..+8
+  if (interior) {
+    // write: thread covers TWO ADJACENT dst rows (c0+cp, c0+cp+1), cols
+    // r0+lc8..+8 — the adjacent pair turns the scattered per-element
+    // LDS reads into 4-B paired reads (half the ds_read instructions of
+    // the old one-col-per-pass scheme)
+    const int cp = (threadIdx.x / 8) * 2;
+    bf16x8 v0, v1;
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int c = lr + p * 32;     // source col == dst row offset
-    bf16x8 v;
+    for (int e = 0; e < 8; ++e) {
+      const unsigned int pr = *reinterpret_cast<const unsigned int*>(
+          &tile[lc8 + e][cp]);
+      v0[e] = __builtin_bit_cast(bf16, (unsigned short)(pr & 0xffffu));
+      v1[e] = __builtin_bit_cast(bf16, (unsigned short)(pr >> 16));
+    }
+    *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + cp) * ldd + r0 + lc8) =
+        v0;
+    *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + cp + 1) * ldd + r0 +
+                               lc8) = v1;
+  } else {
 #pragma unroll
-    for (int e = 0; e < 8; ++e) v[e] = tile[lc8 + e][c];
-    if (interior) {
-      *reinterpret_cast<bf16x8*>(dst + (int64_t)(c0 + c) * ldd + r0 + lc8) = v;
-    } else if (c0 + c < C) {
+    for (int p = 0; p < 2; ++p) {
+      const int c = lr + p * 32;   // source col == dst row offset
+      bf16x8 v;
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        if (r0 + lc8 + e < R)
-          dst[(int64_t)(c0 + c) * ldd + r0 + lc8 + e] = v[e];
+      for (int e = 0; e < 8; ++e) v[e] = tile[lc8 + e][c];
+      if (c0 + c < C) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (r0 + lc8 + e < R)
+            dst[(int64_t)(c0 + c) * ldd + r0 + lc8 + e] = v[e];
+      }
     }
   }
 }
