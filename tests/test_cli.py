@@ -81,3 +81,24 @@ def test_bench_gpus_flag_is_authoritative():
     r = run_cli(["bench.py", "--gpus", "2", "--steps", "1", "--warmup", "0"])
     assert r.returncode != 0
     assert "world size" in (r.stderr + r.stdout)
+
+
+@pytest.mark.timeout(300)
+def test_bench_torchrun_dp2_contract():
+    """The EXACT launch the driver uses for the scaling bench, world=2
+    (CPU/gloo here; RCCL on GPU nodes): one JSON line from rank 0 with
+    the whole-job aggregate and the dp2 config."""
+    import json as _json
+
+    r = run_cli(["-m", "torch.distributed.run", "--nnodes=1",
+                 "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+                 "--master-port", "29575", "bench.py", "--gpus", "2",
+                 "--steps", "2", "--warmup", "1", "--hidden_size", "64",
+                 "--vocab", "200", "--batch_size", "4", "--seq_length", "8"])
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout[-1500:]  # rank 0 only
+    d = _json.loads(lines[0])
+    assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 8  # world * per-rank B
+    assert d["value"] > 0 and d["scaling"] == "weak"
