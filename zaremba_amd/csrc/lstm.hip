@@ -598,6 +598,254 @@ __global__ __launch_bounds__(NTHR) void smallm_fused_bwd_kernel(
   BPROF_EXIT(bp5);
 }
 
+// ---------------------------------------------------------------------------
+// 2-step-batched fused backward (ZAREMBA_AMD_BWD_BATCH2=1): one launch
+// runs hop[t]+dgate[t-1], a padded grid barrier + agent acquire, then
+// hop[t-1]+dgate[t-2] — halving the launch count (the launch-boundary
+// gap measures ~2 us/launch). The carried cell grad dc stays in a
+// REGISTER between the two in-launch steps (the (row, col) partition is
+// identical), and the mid-launch dgate pack is published WRITE-THROUGH
+// so the second hop's plain loads (after the acquire) see it.
+// ---------------------------------------------------------------------------
+typedef __attribute__((address_space(1))) unsigned short gau16;
+DEV_INLINE void store_wt_bf16(bf16* p, bf16 v) {
+  __hip_atomic_store((gau16*)(uintptr_t)p,
+                     __builtin_bit_cast(unsigned short, v), ZRLX_AGENT);
+}
+
+// padded two-level grid barrier — a copy of the proven forward barrier
+// (lstm_persistent.hip xcd_grid_barrier): pstate words [0, 513),
+// monotonic generations, every hot word on its own 128-B line.
+DEV_INLINE bool fused2_grid_barrier(unsigned int* pstate, int NBLK,
+                                    unsigned int gen,
+                                    unsigned int* abort_flag) {
+  __shared__ int ok_s2;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    ok_s2 = 1;
+    gau32* st = (gau32*)(uintptr_t)pstate;
+    const int grp = blockIdx.x & 7;
+    const int ngroups = NBLK < 8 ? NBLK : 8;
+    const int nbg = (NBLK - grp + 7) / 8;
+    unsigned int t0 = __hip_atomic_fetch_add(&st[grp * 32], 1u, ZRLX_AGENT);
+    if (t0 == gen * (unsigned int)nbg - 1) {
+      unsigned int tt = __hip_atomic_fetch_add(&st[256], 1u, ZRLX_AGENT);
+      if (tt == gen * (unsigned int)ngroups - 1)
+        for (int x = 0; x < 8; ++x)
+          __hip_atomic_store(&st[288 + x * 32], gen, ZRLX_AGENT);
+    }
+    unsigned int spins = 0;
+    while (__hip_atomic_load(&st[288 + grp * 32], ZRLX_AGENT) < gen) {
+      __builtin_amdgcn_s_sleep(2);
+      if (++spins > 20000000u) {
+        atomicOr(abort_flag, 1u);
+        ok_s2 = 0;
+        break;
+      }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  return ok_s2 != 0;
+}
+
+template <int MAXB, int NTHR>
+__global__ __launch_bounds__(NTHR) void smallm_fused_bwd2_kernel(
+    const bf16* __restrict__ A_pack,  // prev launch's packed dgates
+    const bf16* __restrict__ W_pack,
+    float* __restrict__ P,            // [2][M*N] partials (reused per half)
+    const bf16* __restrict__ dY,      // base [T, M, N]
+    float* __restrict__ dc,           // [M, N] carried grad (launch hand-off)
+    const bf16* __restrict__ rec,     // base [T] block records
+    bf16* __restrict__ dG,            // base [T, M, K]
+    bf16* __restrict__ pack_mid,      // dgate[t-1] pack (in-launch exchange)
+    bf16* __restrict__ pack_out,      // dgate[t-2] pack (next launch reads)
+    unsigned int* __restrict__ flags,
+    unsigned int* __restrict__ pstate,
+    unsigned int* __restrict__ abort_flag,
+    int M, int N, int K, int HSp, int t, unsigned int step1,
+    unsigned int gen, int64_t rstep) {
+  constexpr int NW = NTHR / 64;
+  __shared__ float red[NW * MAXB * 16];
+  __shared__ int ok_s;
+
+  const int nbn = (N + 15) / 16;
+  const int nb = blockIdx.x / 2;  // partner-interleaved (see fused1)
+  const int n0 = nb * 16;
+  const int sk = blockIdx.x % 2;
+  const int w = wave_id();
+  const int l = lane_id();
+  const int lm = l & 15;
+  const int KS = (K + 31) / 32;
+  const int KH = (KS + 1) / 2;
+  const int ks0 = sk * KH;
+  const int ks1 = min(ks0 + KH, KS);
+  const int Bh = (M + 1) / 2;
+  const int rb0 = sk * Bh;
+  const int nrows = min(M - rb0, Bh);
+  const int db = rb0 + threadIdx.x / 16;
+  const int dj = n0 + threadIdx.x % 16;
+  const bool dwork = ((int)threadIdx.x < nrows * 16) && (dj < N);
+  const int64_t e = (int64_t)db * N + dj;
+  const int fr0 = (l >> 4) * 4;
+  const bf16x8* pw =
+      reinterpret_cast<const bf16x8*>(W_pack) + (int64_t)nb * KS * 64 + l;
+
+  const bf16* asrc[2] = {A_pack, pack_mid};
+  bf16* pdst[2] = {pack_mid, pack_out};
+  const int ts[2] = {t - 1, t - 2};
+  float dc_carry = 0.f;
+
+#pragma unroll
+  for (int h2 = 0; h2 < 2; ++h2) {
+    // ---- hop MFMA over asrc[h2] (same body as fused1) ------------------
+    f32x4 acc0 = {}, acc1 = {};
+    {
+      const bf16x8* pa = reinterpret_cast<const bf16x8*>(asrc[h2]) + l;
+      const int nown = (ks1 - ks0 - w + NW - 1) / NW;
+      int i = 0;
+      for (; i + 8 <= nown; i += 8) {
+        bf16x8 a0v[8], a1v[8], bwv[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int ks = ks0 + w + NW * (i + u);
+          a0v[u] = pa[ks * 128];
+          a1v[u] = pa[ks * 128 + 64];
+          bwv[u] = pw[ks * 64];
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          acc0 = mfma_16x16x32_bf16(a0v[u], bwv[u], acc0);
+          acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
+        }
+      }
+      for (; i < nown; ++i) {
+        const int ks = ks0 + w + NW * i;
+        bf16x8 a0v = pa[ks * 128];
+        bf16x8 a1v = pa[ks * 128 + 64];
+        bf16x8 bwv = pw[ks * 64];
+        acc0 = mfma_16x16x32_bf16(a0v, bwv, acc0);
+        acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
+      }
+    }
+    // ---- prefetch this half's dgate inputs -----------------------------
+    float p_i = 0.f, p_f = 0.f, p_o = 0.f, p_n = 0.f, p_tc = 0.f,
+          p_cprev = 0.f, p_dy = 0.f, p_dc = 0.f;
+    if (dwork) {
+      const int blk = dj / HSp, jr = dj % HSp;
+      const bf16* r =
+          rec + (int64_t)ts[h2] * rstep + (((int64_t)blk * M + db) * 6) * HSp;
+      p_i = bf2f(r[0 * HSp + jr]);
+      p_f = bf2f(r[1 * HSp + jr]);
+      p_o = bf2f(r[2 * HSp + jr]);
+      p_n = bf2f(r[3 * HSp + jr]);
+      p_tc = bf2f(r[4 * HSp + jr]);
+      p_cprev = bf2f(r[5 * HSp + jr]);
+      p_dy = bf2f(dY[(int64_t)ts[h2] * M * N + e]);
+      p_dc = (h2 == 0) ? dc[e] : dc_carry;
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      red[(w * MAXB + fr0 + r) * 16 + lm] = acc0[r];
+      red[(w * MAXB + 16 + fr0 + r) * 16 + lm] = acc1[r];
+    }
+    __syncthreads();
+    // ---- publish this K-half's partial (write-through) -----------------
+    float* mine = P + (int64_t)sk * M * N;
+    for (int idx = threadIdx.x; idx < M * 16; idx += NTHR) {
+      const int b = idx / 16, jj = idx % 16;
+      if (n0 + jj >= N) continue;
+      float v = 0.f;
+#pragma unroll
+      for (int ww = 0; ww < NW; ++ww) v += red[(ww * MAXB + b) * 16 + jj];
+      store_wt_f32(mine + (int64_t)b * N + n0 + jj, v);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    // ---- pair arrive + spin --------------------------------------------
+    if (threadIdx.x == 0) {
+      ok_s = 1;
+      gau32* f = (gau32*)(uintptr_t)(flags + nb);
+      __hip_atomic_fetch_add(f, 1u, ZRLX_AGENT);
+      unsigned int spins = 0;
+      const unsigned int want = 2u * (step1 + (unsigned int)h2);
+      while (__hip_atomic_load(f, ZRLX_AGENT) < want) {
+        __builtin_amdgcn_s_sleep(2);
+        if (++spins > 20000000u) {
+          atomicOr(abort_flag, 1u);
+          ok_s = 0;
+          break;
+        }
+      }
+    }
+    __syncthreads();
+    if (!ok_s) return;
+    // ---- dgate for this half's step ------------------------------------
+    if (dwork) {
+      const int b = db, j = dj, jj = dj - n0;
+      float vo = 0.f;
+#pragma unroll
+      for (int ww = 0; ww < NW; ++ww) vo += red[(ww * MAXB + b) * 16 + jj];
+      const float ps = load_wt_f32(P + (int64_t)(sk ^ 1) * M * N + e);
+      const float dh = p_dy + (vo + ps);
+      const float do_ = dh * p_tc;
+      const float dct = p_dc + dh * p_o * (1.f - p_tc * p_tc);
+      const float di = dct * p_n;
+      const float df = dct * p_cprev;
+      const float dn = dct * p_i;
+      const bf16 v[4] = {f2bf(di * p_i * (1.f - p_i)),
+                         f2bf(df * p_f * (1.f - p_f)),
+                         f2bf(do_ * p_o * (1.f - p_o)),
+                         f2bf(dn * (1.f - p_n * p_n))};
+      bf16* dGt = dG + (int64_t)ts[h2] * M * K;
+      bf16* pk = pdst[h2];
+      const int64_t gbase = (int64_t)b * K + j;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        const int k = g * N + j;
+        dGt[gbase + (int64_t)g * N] = v[g];
+        const int ks = k / 32, sub = k % 32;
+        const int pl = (b & 15) + 16 * (sub / 8);
+        bf16* pp = pk + (((int64_t)ks * 2 + b / 16) * 64 + pl) * 8 + sub % 8;
+        if (h2 == 0)
+          store_wt_bf16(pp, v[g]);  // in-launch consumer after the barrier
+        else
+          *pp = v[g];               // next-launch consumer (kernel boundary)
+      }
+      if (h2 == 0)
+        dc_carry = dct * p_f;
+      else
+        dc[e] = dct * p_f;
+    }
+    // ---- between halves: all dgate[t-1] packs visible everywhere -------
+    if (h2 == 0) {
+      if (!fused2_grid_barrier(pstate, nbn * 2, gen, abort_flag)) return;
+    }
+  }
+}
+
+void launch_smallm_fused_bwd2(const bf16* A_pack, const bf16* W_pack,
+                              float* P, const bf16* dY, float* dc,
+                              const bf16* rec, bf16* dG, bf16* pack_mid,
+                              bf16* pack_out, unsigned int* flags,
+                              unsigned int* pstate,
+                              unsigned int* abort_flag, int M, int N, int K,
+                              int HSp, int t, unsigned int step1,
+                              unsigned int gen, int64_t rstep, int nthreads,
+                              hipStream_t stream) {
+#define ZAMD_FB2_LAUNCH(NT_)                                                \
+  hipLaunchKernelGGL((smallm_fused_bwd2_kernel<32, NT_>),                   \
+                     dim3(cdiv(N, 16) * 2), dim3(NT_), 0, stream, A_pack,   \
+                     W_pack, P, dY, dc, rec, dG, pack_mid, pack_out, flags, \
+                     pstate, abort_flag, M, N, K, HSp, t, step1, gen, rstep)
+  if (nthreads == 1024) ZAMD_FB2_LAUNCH(1024);
+  else if (nthreads == 768) ZAMD_FB2_LAUNCH(768);
+  else if (nthreads == 512) ZAMD_FB2_LAUNCH(512);
+  else ZAMD_FB2_LAUNCH(256);
+#undef ZAMD_FB2_LAUNCH
+}
+
 void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
                              float* P, const bf16* dy, float* dc,
                              const bf16* rec, bf16* dG, bf16* dG_pack_out,
