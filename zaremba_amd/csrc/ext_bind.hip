@@ -171,6 +171,10 @@ static bool g_use_fused_bwd = true;
 // 1 block/CU comes from occupancy, and the bounded-spin abort flag
 // catches any scheduler that breaks it.
 static int g_bwd_ksplit = 2;
+// 2-step-batched fused backward (grid barrier + acquire between the two
+// in-launch steps; see smallm_fused_bwd2_kernel). Default off pending
+// the same-box A/B.
+static bool g_bwd_batch2 = false;
 // Threads per fused-bwd block (256 = 4 waves, 512 = 8): 8-wave blocks
 // double the in-flight load streams per CU at the same 1-block/CU grid.
 // Measured (profiles/s3_census_w*.txt): hop-mfma 6.32 -> 3.91 us with
@@ -316,6 +320,28 @@ static void lstm_seq_bwd_body(const bf16* dY, const bf16* gates,
     launch_lstm_cell_bwd_elt(dY + (T - 1) * hstep, nullptr, dh2, dc,
                              rec + (T - 1) * rstep, dG + (T - 1) * gstep,
                              dG_pack, B, H, HSp, stream);
+    if (g_bwd_batch2 && g_bwd_ksplit == 2 && T >= 3) {
+      // 2-step-batched variant: barrier state in hgran words [0, 513)
+      // (shared with the fwd barrier — sequential, re-zeroed here)
+      unsigned int* pstate = reinterpret_cast<unsigned int*>(hgran);
+      HIP_CHECK(hipMemsetAsync(pstate, 0, 513 * sizeof(unsigned int),
+                               stream));
+      int i = 0, t = T - 1;
+      for (; t >= 2; t -= 2, ++i) {
+        launch_smallm_fused_bwd2(
+            dG_pack, WT_pack, dh_rec, dY, dc, rec, dG,
+            dG_pack + pstride, dG_pack, flags, pstate, abort_flag, B, H,
+            4 * H, HSp, t, (unsigned int)(2 * i + 1), (unsigned int)(i + 1),
+            rstep, g_bwd_threads, stream);
+      }
+      if (t == 1) {  // even T: one leftover single-step launch
+        launch_smallm_fused_bwd(dG_pack, WT_pack, dh_rec, dY, dc, rec, dG,
+                                dG_pack + pstride, flags, abort_flag, B, H,
+                                4 * H, HSp, (unsigned int)(2 * i + 1), 2,
+                                g_bwd_threads, stream);
+      }
+      return;
+    }
     for (int t = T - 1; t >= 1; --t) {
       const int i = T - 1 - t;
       bf16* rd = dG_pack + (i & 1) * pstride;
@@ -413,6 +439,7 @@ static void set_bwd_ksplit(int v) {
   TORCH_CHECK(v == 2 || v == 4, "bwd ksplit must be 2 or 4");
   g_bwd_ksplit = v;
 }
+static void set_bwd_batch2(bool v) { g_bwd_batch2 = v; }
 static void set_bwd_threads(int v) {
   TORCH_CHECK(v == 256 || v == 512 || v == 768 || v == 1024,
               "bwd threads must be 256, 512, 768 or 1024");
@@ -633,6 +660,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_use_fused_bwd", &zamd::set_use_fused_bwd);
   m.def("set_bwd_ksplit", &zamd::set_bwd_ksplit);
   m.def("set_bwd_threads", &zamd::set_bwd_threads);
+  m.def("set_bwd_batch2", &zamd::set_bwd_batch2);
   m.def("fused_bwd_active", &zamd::fused_bwd_ok);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

@@ -41,6 +41,15 @@ void launch_lstm_cell_bwd_elt(const bf16* dy, const float* dh_rec,
 void launch_smallm_packed_nt(const bf16* A_pack, const bf16* W_pack,
                              float* C, float* C2, int M, int N, int K,
                              hipStream_t stream);
+void launch_smallm_fused_bwd2(const bf16* A_pack, const bf16* W_pack,
+                              float* P, const bf16* dY, float* dc,
+                              const bf16* rec, bf16* dG, bf16* pack_mid,
+                              bf16* pack_out, unsigned int* flags,
+                              unsigned int* pstate,
+                              unsigned int* abort_flag, int M, int N, int K,
+                              int HSp, int t, unsigned int step1,
+                              unsigned int gen, int64_t rstep, int nthreads,
+                              hipStream_t stream);
 void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
                              float* P, const bf16* dy, float* dc,
                              const bf16* rec, bf16* dG, bf16* dG_pack_out,

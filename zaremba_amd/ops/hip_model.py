@@ -178,6 +178,8 @@ class HipModel:
             int(os.environ.get("ZAREMBA_AMD_BWD_KSPLIT", "2")))
         self.e.set_bwd_threads(
             int(os.environ.get("ZAREMBA_AMD_BWD_WAVES", "16")) * 64)
+        if os.environ.get("ZAREMBA_AMD_BWD_BATCH2", "0") == "1":
+            self.e.set_bwd_batch2(True)
         self.compute_dtype = torch.bfloat16
         dev = self.device
         self.emb_W = torch.empty_like(model.embed.W, dtype=torch.bfloat16,
