@@ -205,19 +205,34 @@ def test_lstm_seq_fwd_matches_step_loop(ext, H):
 
 def test_lstm_seq_fwd_persistent_matches_per_step(ext):
     """The persistent one-launch path and the per-step fallback must agree
-    on the h trajectory and carried state (same math, same MFMA order)."""
+    on the h trajectory and carried state. At 4 waves the MFMA order is
+    identical -> BITWISE equality (the race-detection oracle); the
+    default 8-wave path splits K by granule parity (different f32
+    association) -> allclose."""
     T, B, H = 7, 20, 1500
-    gx1, _, h1, c1, _ = _run_seq_fwd(ext, T, B, H, seed=11)
-    h1 = h1.clone()
-    cT1 = c1[T].clone()
+    # --- 4-wave persistent: bitwise vs the per-step fallback
+    ext.set_fwd_threads(256)
+    try:
+        gx1, _, h1, c1, _ = _run_seq_fwd(ext, T, B, H, seed=11)
+        h1 = h1.clone()
+        cT1 = c1[T].clone()
+    finally:
+        ext.set_fwd_threads(512)
     ext.set_use_persistent(False)
     try:
         gx2, _, h2, c2, _ = _run_seq_fwd(ext, T, B, H, seed=11)
+        h2 = h2.clone()
+        cT2 = c2[T].clone()
     finally:
         ext.set_use_persistent(True)
     assert torch.equal(gx1, gx2)
     assert torch.equal(h1, h2)
-    assert torch.allclose(cT1, c2[T], atol=1e-6)
+    assert torch.allclose(cT1, cT2, atol=1e-6)
+    # --- default 8-wave persistent (K-parity split): allclose
+    gx3, _, h3, c3, _ = _run_seq_fwd(ext, T, B, H, seed=11)
+    assert torch.equal(gx3, gx2)
+    assert torch.allclose(h3.float(), h2.float(), atol=3e-2, rtol=1e-2)
+    assert torch.allclose(c3[T], cT2, atol=1e-3, rtol=1e-3)
 
 
 @pytest.mark.parametrize("B,ksplit,waves", [(20, 2, 4), (7, 2, 4),

@@ -440,6 +440,10 @@ static void set_bwd_ksplit(int v) {
   g_bwd_ksplit = v;
 }
 static void set_bwd_batch2(bool v) { g_bwd_batch2 = v; }
+static void set_fwd_threads(int v) {
+  TORCH_CHECK(v == 256 || v == 512, "fwd threads must be 256 or 512");
+  set_fwd_threads_impl(v);
+}
 static void set_bwd_threads(int v) {
   TORCH_CHECK(v == 256 || v == 512 || v == 768 || v == 1024,
               "bwd threads must be 256, 512, 768 or 1024");
@@ -661,6 +665,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_bwd_ksplit", &zamd::set_bwd_ksplit);
   m.def("set_bwd_threads", &zamd::set_bwd_threads);
   m.def("set_bwd_batch2", &zamd::set_bwd_batch2);
+  m.def("set_fwd_threads", &zamd::set_fwd_threads);
   m.def("fused_bwd_active", &zamd::fused_bwd_ok);
   m.def("clear_graphs", &zamd::clear_graphs);
 }

@@ -59,6 +59,7 @@ void launch_smallm_fused_bwd(const bf16* A_pack, const bf16* W_pack,
 
 // lstm_persistent.hip — one launch for a whole layer unroll
 int persistent_hs(int H);
+void set_fwd_threads_impl(int v);
 size_t persistent_fwd_lds(int B, int H);
 void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
                                 float* c_all, bf16* rec,

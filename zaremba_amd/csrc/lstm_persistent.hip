@@ -42,6 +42,17 @@ typedef __attribute__((address_space(1))) unsigned int gu32;
 constexpr int PCELL_THREADS = 256;
 constexpr int REC_CH = 6;  // i,f,o,n,tanh_c_new,c_prev channels
 
+// Forward threads/block: 512 (8 waves: memory phases get 2x the streams
+// AND the gate MFMA splits across 2 waves per gate by K-parity;
+// measured best) or 256 (4 waves, serial K per gate — the bitwise
+// persistent-vs-per-step oracle path). env ZAREMBA_AMD_FWD_WAVES
+// initializes; set_fwd_threads() overrides at runtime (tests).
+int g_fwd_threads = [] {
+  const char* e = getenv("ZAREMBA_AMD_FWD_WAVES");
+  return e ? atoi(e) * 64 : 512;
+}();
+void set_fwd_threads_impl(int v) { g_fwd_threads = v; }
+
 // HS must be even; NB = ceil(H/HS) <= 250 so the grid is co-resident.
 // Prefer HS=8 (NB=188 at H=1500): the grid barrier is the forward's
 // dominant per-step cost and scales with workgroup count (census probe:
@@ -179,7 +190,15 @@ __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
   const int NB = (H + HS - 1) / HS;
 
   const int j0 = blockIdx.x * HS;
-  const int g = wave_id();
+  // 8-wave blocks split each gate's K reduction across 2 waves by
+  // granule PARITY (wave w -> gate w&3, K-parity w>>2): even granules
+  // on waves 0-3, odd on 4-7, combined in a fixed two-phase gbuf
+  // exchange (no atomics, no extra LDS — deterministic, but a
+  // different f32 association than the serial 4-wave path, so the
+  // bitwise persistent-vs-per-step oracle runs at 4 waves).
+  constexpr int NWG = NTHR / 256;  // K-parity ways (1 or 2)
+  const int g = wave_id() & 3;
+  const int kh = wave_id() >> 2;
   const int l = lane_id();
   const int lm = l & 15;
   const int t_ = threadIdx.x;
@@ -316,20 +335,18 @@ __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
     PROF_ACC(2, pt2, pt3);  // stage half A + issue half B
 #endif
 
-    // ---- gate MFMA reduction (wave g -> gate g), half A ------------------
-    // waves >= 4 (NTHR > 256) skip the MFMA: they exist for the memory
-    // phases; all barriers below are outside this guard
+    // ---- gate MFMA reduction (wave -> gate g, K-parity kh) ---------------
     f32x4 acc0 = {}, acc1 = {};
     const bf16* pw = Ws + (int64_t)(g * HS + wc) * KPAD;
     const bf16* pa0 = hs + (int64_t)a0r * KPAD;
     const bf16* pa1 = hs + (int64_t)a1r * KPAD;
     auto mfma_range = [&](int ks_lo, int ks_hi) {
-      int ks = ks_lo;
-      for (; ks + 4 <= ks_hi; ks += 4) {
+      int ks = ks_lo + kh;
+      for (; ks + 3 * NWG < ks_hi; ks += 4 * NWG) {
         bf16x8 a0v[4], a1v[4], bwv[4];
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
-          const int k = (ks + u) * 32 + lk;
+          const int k = (ks + u * NWG) * 32 + lk;
           a0v[u] = *reinterpret_cast<const bf16x8*>(pa0 + k);
           a1v[u] = *reinterpret_cast<const bf16x8*>(pa1 + k);
           bwv[u] = *reinterpret_cast<const bf16x8*>(pw + k);
@@ -340,7 +357,7 @@ __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
           acc1 = mfma_16x16x32_bf16(a1v[u], bwv[u], acc1);
         }
       }
-      for (; ks < ks_hi; ++ks) {
+      for (; ks < ks_hi; ks += NWG) {
         const int k = ks * 32 + lk;
         bf16x8 a0v = *reinterpret_cast<const bf16x8*>(pa0 + k);
         bf16x8 a1v = *reinterpret_cast<const bf16x8*>(pa1 + k);
@@ -349,7 +366,7 @@ __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
         acc1 = mfma_16x16x32_bf16(a1v, bwv, acc1);
       }
     };
-    if (g < 4) mfma_range(0, ksA);
+    mfma_range(0, ksA);
     // ---- land half B in LDS, then finish the reduction -------------------
     if (vB > 0) {
       {
@@ -372,14 +389,26 @@ __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
         }
       }
       __syncthreads();  // half B visible
-      if (g < 4) mfma_range(ksA, KS);
+      mfma_range(ksA, KS);
     }
-    if (g < 4) {
+    // two-phase deterministic combine: even-K waves write, odd-K add
+    if (kh == 0) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         if (fr0 + r < B) gbuf[(g * B + fr0 + r) * 16 + lm] = acc0[r];
         if (16 + fr0 + r < B)
           gbuf[(g * B + 16 + fr0 + r) * 16 + lm] = acc1[r];
+      }
+    }
+    if (NWG == 2) {
+      __syncthreads();
+      if (kh == 1) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          if (fr0 + r < B) gbuf[(g * B + fr0 + r) * 16 + lm] += acc0[r];
+          if (16 + fr0 + r < B)
+            gbuf[(g * B + 16 + fr0 + r) * 16 + lm] += acc1[r];
+        }
       }
     }
     __syncthreads();
@@ -449,13 +478,7 @@ void launch_lstm_persistent_fwd(const bf16* gx, const bf16* W_h, bf16* h_all,
   const int HS = persistent_hs(H);
   const int NB = cdiv(H, HS);
   size_t lds = persistent_fwd_lds(B, H);
-  // ZAREMBA_AMD_FWD_WAVES: 8 (default) or 4 — 8-wave blocks give the
-  // memory phases twice the streams at unchanged MFMA/LDS layout
-  // (measured 329.0K vs 324.7K tokens/s same-box; s18 censuses).
-  static const int nthr = [] {
-    const char* e = getenv("ZAREMBA_AMD_FWD_WAVES");
-    return e ? atoi(e) * 64 : 512;
-  }();
+  const int nthr = g_fwd_threads;
   if (nthr == 512) {
     hipLaunchKernelGGL((lstm_persistent_fwd_kernel<32, 512>), dim3(NB),
                        dim3(512), lds, stream, gx, W_h, h_all, c_all,
