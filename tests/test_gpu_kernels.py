@@ -34,7 +34,11 @@ def rel_err(a, b):
 # GEMM
 # ---------------------------------------------------------------------------
 @pytest.mark.parametrize("M,N,K", [(128, 128, 64), (700, 6000, 1500),
-                                   (700, 10000, 1500), (37, 129, 65)])
+                                   (700, 10000, 1500), (37, 129, 65),
+                                   # per-shape tile heuristic paths:
+                                   (6000, 1500, 768),    # 128x64 dW family
+                                   (10000, 1500, 704),   # >8192-M -> 128^2
+                                   (448, 6016, 1536)])   # wide-N -> 256x128
 def test_gemm_nt(ext, M, N, K):
     torch.manual_seed(0)
     A = torch.randn(M, K, device=dev(), dtype=torch.bfloat16)
