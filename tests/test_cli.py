@@ -102,3 +102,24 @@ def test_bench_torchrun_dp2_contract():
     assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 8  # world * per-rank B
     assert d["value"] > 0 and d["scaling"] == "weak"
+
+
+def test_ensemble_jsonl_per_model_suffix(monkeypatch):
+    """Multi-rank ensemble mode must not share one JSONL file across
+    concurrently-training ranks (advisor finding): the path is suffixed
+    per model; single-process mode keeps the user's path."""
+    import importlib
+    import sys as _sys
+    _sys.path.insert(0, REPO)
+    ens = importlib.import_module("ensemble")
+    from zaremba_amd.parallel import dist as zdist
+
+    class A:
+        jsonl = "/tmp/run.jsonl"
+
+    monkeypatch.setattr(zdist, "world_size", lambda: 1)
+    assert ens.jsonl_for_model(A, 3) == "/tmp/run.jsonl"
+    monkeypatch.setattr(zdist, "world_size", lambda: 4)
+    assert ens.jsonl_for_model(A, 3) == "/tmp/run.model3.jsonl"
+    A.jsonl = None
+    assert ens.jsonl_for_model(A, 1) is None
