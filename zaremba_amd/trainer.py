@@ -206,6 +206,8 @@ def train(
                 from .checkpoint import save_checkpoint
                 save_checkpoint(save_path, model, epoch=epoch + 1, lr=lr)
     tst_perp = perplexity(tst, model, batch_size)
+    if model._hip_model is not None:
+        model._hip_model.check_aborts()
     if is_rank0:
         print("Test set perplexity : {:.3f}".format(tst_perp))
         print("Training is over." if model_num is None else
