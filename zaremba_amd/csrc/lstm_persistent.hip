@@ -165,9 +165,11 @@ __device__ unsigned long long g_fwd_prof[256 * 8];
 // ===========================================================================
 // Forward
 // ===========================================================================
-// NTHR: 256 = 4 waves; 512 = 8 waves where the EXTRA waves only help
-// the memory phases (h staging, publish, W preload) — the gate MFMA
-// stays on waves 0-3 so gbuf/LDS layout is unchanged.
+// NTHR: 256 = 4 waves (one wave per gate, serial K — the bitwise
+// oracle path); 512 = 8 waves: the memory phases (h staging, publish,
+// W preload) get 2x the streams AND each gate's K reduction is split
+// across 2 waves by granule parity with a deterministic two-phase gbuf
+// combine (same LDS layout either way).
 template <int MAXB, int NTHR>
 __global__ __launch_bounds__(NTHR) void lstm_persistent_fwd_kernel(
     const bf16* __restrict__ gx,     // [T, B, 4H]
