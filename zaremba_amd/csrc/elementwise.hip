@@ -403,19 +403,29 @@ __global__ void norm2_mt_kernel(const int64_t* __restrict__ desc,
   __shared__ float scratch[8];
   const float* g = reinterpret_cast<const float*>(desc[blockIdx.x * 2]);
   const int n = (int)desc[blockIdx.x * 2 + 1];
-  float acc = 0.f;
+  // 2 independent accumulator chains (the single-chain loop was a
+  // serial dependent-add per strided float4 load — the same latency
+  // pathology the colsum kernels had)
+  float a0 = 0.f, a1 = 0.f;
   const int n4 = n / 4;
   const float4* g4 = reinterpret_cast<const float4*>(g);
-  for (int i = threadIdx.x; i < n4; i += blockDim.x) {
+  int i = threadIdx.x;
+  for (; i + (int)blockDim.x < n4; i += 2 * blockDim.x) {
     float4 v = g4[i];
-    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    float4 u = g4[i + blockDim.x];
+    a0 += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    a1 += u.x * u.x + u.y * u.y + u.z * u.z + u.w * u.w;
+  }
+  if (i < n4) {
+    float4 v = g4[i];
+    a0 += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
   }
   if (threadIdx.x < (n & 3)) {
     float v = g[n4 * 4 + threadIdx.x];
-    acc += v * v;
+    a0 += v * v;
   }
-  acc = block_reduce(acc, scratch, 0.f,
-                     [] __device__(float a, float b) { return a + b; });
+  float acc = block_reduce(a0 + a1, scratch, 0.f,
+                           [] __device__(float a, float b) { return a + b; });
   if (threadIdx.x == 0) atomicAdd(accum, acc);
 }
 
