@@ -283,7 +283,22 @@ __global__ void lsm_nll_bwd_kernel(const float* __restrict__ scores,
   float* d = dscores + (int64_t)row * V;
   float l = lse[row];
   int64_t yy = y[row];
-  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+  // vec4 main body (V4 = V/4*4); per-lane float4 keeps the wave's
+  // access one contiguous 1 KB line
+  const int V4 = V / 4;
+  const float4* s4 = reinterpret_cast<const float4*>(s);
+  float4* d4 = reinterpret_cast<float4*>(d);
+  for (int q = threadIdx.x; q < V4; q += blockDim.x) {
+    float4 v = s4[q];
+    const int base = q * 4;
+    float4 o;
+    o.x = (__expf(v.x - l) - (base + 0 == yy ? 1.f : 0.f)) * g;
+    o.y = (__expf(v.y - l) - (base + 1 == yy ? 1.f : 0.f)) * g;
+    o.z = (__expf(v.z - l) - (base + 2 == yy ? 1.f : 0.f)) * g;
+    o.w = (__expf(v.w - l) - (base + 3 == yy ? 1.f : 0.f)) * g;
+    d4[q] = o;
+  }
+  for (int v = V4 * 4 + threadIdx.x; v < V; v += blockDim.x) {
     float p = __expf(s[v] - l);
     d[v] = (p - (v == yy ? 1.f : 0.f)) * g;
   }
