@@ -199,7 +199,8 @@ __global__ void lsm_nll_fwd_kernel(const float* __restrict__ scores,
   // dependent update per strided load — latency-bound at ~1.2 TB/s
   // over the [700,10000] scores row; vec4 loads quarter the
   // latency-per-element and the chain pair overlaps them.
-  const int V4 = V / 4;
+  // rows are float4-aligned only when V % 4 == 0 (s + row*V)
+  const int V4 = (V & 3) == 0 ? V / 4 : 0;
   const int half4 = (V4 >= 2 * (int)blockDim.x)
                         ? (int)(((V4 + 2 * blockDim.x - 1) /
                                  (2 * blockDim.x)) * blockDim.x)
@@ -294,9 +295,10 @@ __global__ void lsm_nll_bwd_kernel(const float* __restrict__ scores,
   float* d = dscores + (int64_t)row * V;
   float l = lse[row];
   int64_t yy = y[row];
-  // vec4 main body (V4 = V/4*4); per-lane float4 keeps the wave's
-  // access one contiguous 1 KB line
-  const int V4 = V / 4;
+  // vec4 main body; per-lane float4 keeps the wave's access one
+  // contiguous 1 KB line. Rows are float4-aligned only when V % 4 == 0
+  // (s + row*V) — otherwise the scalar tail loop covers everything.
+  const int V4 = (V & 3) == 0 ? V / 4 : 0;
   const float4* s4 = reinterpret_cast<const float4*>(s);
   float4* d4 = reinterpret_cast<float4*>(d);
   for (int q = threadIdx.x; q < V4; q += blockDim.x) {
@@ -336,16 +338,25 @@ __global__ void softmax_acc_kernel(const float* __restrict__ scores,
   int row = blockIdx.x;
   if (row >= N) return;
   const float* s = scores + (int64_t)row * V;
+  // vec4 online chain (see lsm_nll_fwd: the scalar strided loop was
+  // load-latency bound)
   float m = -INFINITY, a = 0.f;
-  for (int v = threadIdx.x; v < V; v += blockDim.x) {
-    float x = s[v];
+  auto upd = [&](float x) {
     if (x > m) {
       a = a * __expf(m - x) + 1.f;
       m = x;
     } else {
       a += __expf(x - m);
     }
+  };
+  // rows are float4-aligned only when V % 4 == 0 (s + row*V)
+  const int V4 = (V & 3) == 0 ? V / 4 : 0;
+  const float4* s4 = reinterpret_cast<const float4*>(s);
+  for (int q = threadIdx.x; q < V4; q += blockDim.x) {
+    float4 v = s4[q];
+    upd(v.x); upd(v.y); upd(v.z); upd(v.w);
   }
+  for (int v = V4 * 4 + threadIdx.x; v < V; v += blockDim.x) upd(s[v]);
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
     float m2 = __shfl_down(m, off, 64);
@@ -377,7 +388,17 @@ __global__ void softmax_acc_kernel(const float* __restrict__ scores,
   __syncthreads();
   const float M = row_m, invA = 1.f / row_a;
   float* d = acc + (int64_t)row * V;
-  for (int v = threadIdx.x; v < V; v += blockDim.x)
+  float4* d4 = reinterpret_cast<float4*>(d);
+  for (int q = threadIdx.x; q < V4; q += blockDim.x) {
+    float4 v = s4[q];
+    float4 o = d4[q];
+    o.x += __expf(v.x - M) * invA;
+    o.y += __expf(v.y - M) * invA;
+    o.z += __expf(v.z - M) * invA;
+    o.w += __expf(v.w - M) * invA;
+    d4[q] = o;
+  }
+  for (int v = V4 * 4 + threadIdx.x; v < V; v += blockDim.x)
     d[v] += __expf(s[v] - M) * invA;
 }
 
