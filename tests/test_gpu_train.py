@@ -162,3 +162,26 @@ def test_hip_learns_markov_structure():
     # unigram ppl is ~430 here; the conditional structure (optimal 20)
     # must be clearly learned
     assert ppl < 60, ppl
+
+
+def test_checkpoint_restores_cuda_rng(tmp_path):
+    """restore_rng puts the device generators back (round-2 feature;
+    the CPU test can only cover the host stream)."""
+    import torch
+    from zaremba_amd.checkpoint import (load_checkpoint, restore_rng,
+                                        save_checkpoint)
+    from zaremba_amd.models.lstm_lm import Model
+
+    torch.manual_seed(11)
+    torch.cuda.manual_seed_all(11)
+    m = Model(20, 8, 1, dropout=0.0, winit=0.1)
+    torch.rand(32, device=dev())  # advance the device stream
+    state_at_save = torch.cuda.get_rng_state(0).clone()
+    path = str(tmp_path / "rng.pt")
+    save_checkpoint(path, m, epoch=1, lr=1.0)
+    expected_next = torch.rand(16, device=dev())
+    torch.cuda.manual_seed_all(999)
+    torch.rand(7, device=dev())
+    restore_rng(load_checkpoint(path))
+    assert torch.equal(torch.cuda.get_rng_state(0), state_at_save)
+    assert torch.equal(torch.rand(16, device=dev()), expected_next)
