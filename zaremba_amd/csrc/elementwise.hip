@@ -89,19 +89,14 @@ void launch_embedding_bwd_det(const bf16* dY, const int64_t* idx, float* dW,
 // ---------------------------------------------------------------------------
 // K5: dropout (inverted, scale 1/(1-p) at train time)
 // ---------------------------------------------------------------------------
-// The fwd kernel draws its philox offset from a device counter (so graph
-// replays advance the stream), saves the offset it used for backward, and
-// bumps the counter. 4 elements per philox call.
+// The philox offset is a HOST-side counter passed by value: dropout is
+// never hipGraph-captured (capture covers only the C++ LSTM launch
+// trains), so no device counter is needed — this removes a tick kernel
+// + a 1-element alloc per call. The offset is saved in the autograd ctx
+// and the backward regenerates the identical mask. 4 elements/philox.
 __global__ void dropout_fwd_kernel(const bf16* __restrict__ x,
                                    bf16* __restrict__ y, float p,
-                                   uint64_t seed,
-                                   uint64_t* __restrict__ counter,
-                                   uint64_t* __restrict__ saved_offset,
-                                   int64_t n) {
-  __shared__ uint64_t off_s;
-  if (threadIdx.x == 0) off_s = *saved_offset;
-  __syncthreads();
-  uint64_t off = off_s;
+                                   uint64_t seed, uint64_t off, int64_t n) {
   float scale = 1.f / (1.f - p);
   int64_t quad = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t nquads = (n + 3) / 4;
@@ -120,22 +115,9 @@ __global__ void dropout_fwd_kernel(const bf16* __restrict__ x,
   }
 }
 
-// Host-side step: latch the counter into saved_offset, then advance it.
-__global__ void dropout_tick_kernel(uint64_t* counter, uint64_t* saved_offset,
-                                    int64_t nquads) {
-  *saved_offset = *counter;
-  *counter += nquads;
-}
-
 __global__ void dropout_bwd_kernel(const bf16* __restrict__ dy,
                                    bf16* __restrict__ dx, float p,
-                                   uint64_t seed,
-                                   const uint64_t* __restrict__ saved_offset,
-                                   int64_t n) {
-  __shared__ uint64_t off_s;
-  if (threadIdx.x == 0) off_s = *saved_offset;
-  __syncthreads();
-  uint64_t off = off_s;
+                                   uint64_t seed, uint64_t off, int64_t n) {
   float scale = 1.f / (1.f - p);
   int64_t quad = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t nquads = (n + 3) / 4;
@@ -161,19 +143,15 @@ static int dropout_grid(int64_t n) {
 }
 
 void launch_dropout_fwd(const bf16* x, bf16* y, float p, uint64_t seed,
-                        uint64_t* counter, uint64_t* saved_offset, int64_t n,
-                        hipStream_t stream) {
-  hipLaunchKernelGGL(dropout_tick_kernel, dim3(1), dim3(1), 0, stream,
-                     counter, saved_offset, (n + 3) / 4);
+                        uint64_t offset, int64_t n, hipStream_t stream) {
   hipLaunchKernelGGL(dropout_fwd_kernel, dim3(dropout_grid(n)), dim3(256), 0,
-                     stream, x, y, p, seed, counter, saved_offset, n);
+                     stream, x, y, p, seed, offset, n);
 }
 
 void launch_dropout_bwd(const bf16* dy, bf16* dx, float p, uint64_t seed,
-                        const uint64_t* saved_offset, int64_t n,
-                        hipStream_t stream) {
+                        uint64_t offset, int64_t n, hipStream_t stream) {
   hipLaunchKernelGGL(dropout_bwd_kernel, dim3(dropout_grid(n)), dim3(256), 0,
-                     stream, dy, dx, p, seed, saved_offset, n);
+                     stream, dy, dx, p, seed, offset, n);
 }
 
 // ---------------------------------------------------------------------------

@@ -537,19 +537,15 @@ static void embedding_bwd_det(const torch::Tensor& dY,
 }
 
 static void dropout_fwd(const torch::Tensor& x, torch::Tensor& y, double p,
-                        int64_t seed, torch::Tensor& counter,
-                        torch::Tensor& saved_offset) {
+                        int64_t seed, int64_t offset) {
   launch_dropout_fwd(bf_ptr(x), bf_ptr_mut(y), (float)p, (uint64_t)seed,
-                     reinterpret_cast<uint64_t*>(counter.data_ptr()),
-                     reinterpret_cast<uint64_t*>(saved_offset.data_ptr()),
-                     x.numel(), current_stream());
+                     (uint64_t)offset, x.numel(), current_stream());
 }
 
 static void dropout_bwd(const torch::Tensor& dy, torch::Tensor& dx, double p,
-                        int64_t seed, const torch::Tensor& saved_offset) {
+                        int64_t seed, int64_t offset) {
   launch_dropout_bwd(bf_ptr(dy), bf_ptr_mut(dx), (float)p, (uint64_t)seed,
-                     reinterpret_cast<const uint64_t*>(saved_offset.data_ptr()),
-                     dy.numel(), current_stream());
+                     (uint64_t)offset, dy.numel(), current_stream());
 }
 
 static void lsm_nll_fwd(const torch::Tensor& scores, const torch::Tensor& y,

@@ -81,11 +81,9 @@ void launch_embedding_bwd_det(const bf16* dY, const int64_t* idx, float* dW,
 void launch_embedding_bwd(const bf16* dY, const int64_t* idx, float* dW,
                           int N, int H, hipStream_t stream);
 void launch_dropout_fwd(const bf16* x, bf16* y, float p, uint64_t seed,
-                        uint64_t* counter, uint64_t* saved_offset, int64_t n,
-                        hipStream_t stream);
+                        uint64_t offset, int64_t n, hipStream_t stream);
 void launch_dropout_bwd(const bf16* dy, bf16* dx, float p, uint64_t seed,
-                        const uint64_t* saved_offset, int64_t n,
-                        hipStream_t stream);
+                        uint64_t offset, int64_t n, hipStream_t stream);
 void launch_lsm_nll_fwd(const float* scores, const int64_t* y, float* lse,
                         float* loss_accum, int N, int V, hipStream_t stream);
 void launch_lsm_nll_bwd(const float* scores, const float* lse,

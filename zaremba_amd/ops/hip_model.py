@@ -7,7 +7,8 @@ Owns, per model instance:
   * persistent per-layer workspaces (gx, h_all, c_all, gates, dG, ...)
     sized on first use so the C++ LSTM sequence drivers can hipGraph-
     capture the per-timestep launch trains against stable pointers,
-  * the philox dropout counter (device-side, graph-replay safe),
+  * the philox dropout offset counter (host-side: dropout is never
+    hipGraph-captured),
   * the fused clip+SGD step (grad-norm^2 reduce -> per-param update
     rewriting master fp32 + bf16 shadow in one pass -> transposed-shadow
     refresh kernels).
@@ -194,7 +195,7 @@ class HipModel:
         self.fc = _FcRuntime(model.fc, dev)
         self.fc.slack_ptrs = self.slack_ptrs
         self.dropout_seed = secrets.randbits(63)
-        self.dropout_counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.dropout_counter = 0  # host-side philox quad offset
         self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
         self._shadows_fresh = False
         # Side stream for off-critical-path backward work (weight-grad
@@ -219,6 +220,13 @@ class HipModel:
         for rt in self.layers:
             rt.owner = self
         self.fc.owner = self
+
+    def _bump_dropout(self, numel: int) -> int:
+        """Reserve a philox quad range for one dropout call and return
+        its starting offset (host-side stream bookkeeping)."""
+        off = self.dropout_counter
+        self.dropout_counter += (numel + 3) // 4
+        return off
 
     def set_compute_dtype(self, dtype):
         if dtype != torch.bfloat16:
@@ -281,7 +289,7 @@ class HipModel:
         cur = emb.view(T, B, H)
         if training and p > 0:
             cur = DropoutFn.apply(cur, p, self.dropout_seed,
-                                  self.dropout_counter,
+                                  self._bump_dropout(cur.numel()),
                                   [self.slack_buf(("drop", 0), cur.shape, bf)])
         new_states = list(states)
         for i, (rt, layer) in enumerate(zip(self.layers, m.rnns)):
@@ -294,7 +302,7 @@ class HipModel:
             cur = out
             if training and p > 0:
                 cur = DropoutFn.apply(cur, p, self.dropout_seed,
-                                      self.dropout_counter,
+                                      self._bump_dropout(cur.numel()),
                                       [self.slack_buf(("drop", i + 1),
                                                       cur.shape, bf)])
         scores = LinearFn.apply(cur.reshape(T * B, m.hidden_size), m.fc.W,

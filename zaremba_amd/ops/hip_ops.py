@@ -50,29 +50,29 @@ class EmbeddingFn(torch.autograd.Function):
 
 class DropoutFn(torch.autograd.Function):
     """K5: philox inverted dropout; the mask is regenerated in backward
-    from (seed, saved_offset) — no mask tensor stored."""
+    from (seed, offset) — no mask tensor stored. The philox offset is a
+    HOST-side counter (a plain int advanced by the caller): dropout is
+    never hipGraph-captured, so the round-1 device-counter + tick-kernel
+    machinery was pure overhead (a tick launch + a 1-elem alloc per
+    call) and was removed."""
 
     @staticmethod
-    def forward(ctx, x, p, seed, counter, out=None):
+    def forward(ctx, x, p, seed, offset, out=None):
         # out: None or a 1-list [buf] (a non-Tensor holder, so autograd
         # does not treat the reused buffer as an aliased input)
         x = x.contiguous()
         y = torch.empty_like(x) if out is None else out[0].view_as(x)
-        # empty, not zeros: dropout_tick overwrites it before any read
-        # (a zeros() here cost a 5 us fill launch x3 per step)
-        saved_offset = torch.empty(1, dtype=torch.int64, device=x.device)
-        ext().dropout_fwd(x, y, p, seed, counter, saved_offset)
+        ext().dropout_fwd(x, y, p, seed, offset)
         ctx.p = p
         ctx.seed = seed
-        ctx.save_for_backward(saved_offset)
+        ctx.offset = offset
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        (saved_offset,) = ctx.saved_tensors
         dy = dy.contiguous()
         dx = torch.empty_like(dy)
-        ext().dropout_bwd(dy, dx, ctx.p, ctx.seed, saved_offset)
+        ext().dropout_bwd(dy, dx, ctx.p, ctx.seed, ctx.offset)
         return dx, None, None, None, None
 
 
