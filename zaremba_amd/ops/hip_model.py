@@ -17,7 +17,6 @@ Owns, per model instance:
 from __future__ import annotations
 
 import os
-import secrets
 from typing import Optional
 
 import torch
@@ -194,7 +193,13 @@ class HipModel:
             rt.slack_ptrs = self.slack_ptrs
         self.fc = _FcRuntime(model.fc, dev)
         self.fc.slack_ptrs = self.slack_ptrs
-        self.dropout_seed = secrets.randbits(63)
+        # Derive the philox seed from the torch generator so --seed
+        # controls the HIP dropout masks too (round-2 fix: secrets-based
+        # seeding made same-seed GPU runs draw different masks). Resume
+        # restores the torch RNG, so a rebuilt model draws a
+        # deterministic seed there as well.
+        self.dropout_seed = int(torch.randint(
+            0, (1 << 62) - 1, (1,), dtype=torch.int64).item())
         self.dropout_counter = 0  # host-side philox quad offset
         self.norm2 = torch.zeros(1, dtype=torch.float32, device=dev)
         self._shadows_fresh = False
