@@ -98,8 +98,10 @@ def main():
 
     def one_step(x, y):
         nonlocal states
-        model.zero_grad(set_to_none=(dp is None))
-        if dp is not None:
+        # under DP the bucketer zeroes its flat buffers (grads are views)
+        if dp is None:
+            model.zero_grad(set_to_none=True)
+        else:
             dp.zero_grad()
         states = model.detach(states)
         scores, states = model(x, states)

@@ -103,7 +103,13 @@ class GradBucketer:
             b.work = td.all_reduce(b.comm, op=td.ReduceOp.SUM, async_op=True)
 
     def zero_grad(self):
+        """Zero the flat comm buffers (the grads are views into them) and
+        re-arm the countdowns. Callers use THIS instead of
+        model.zero_grad() under DP: one fill per bucket instead of one
+        per parameter (autograd then accumulates into the zeroed
+        views)."""
         for b in self.buckets:
+            b.flat.zero_()
             b.reset()
 
     def finalize_backward(self):

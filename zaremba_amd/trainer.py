@@ -145,10 +145,12 @@ def train(
             lr = lr / factor
         for i, (x, y) in enumerate(trn):
             total_words += x.numel()
-            # set_to_none saves a fill + an accumulate add per param; the DP
-            # bucketer needs stable grad views, so keep zeroing there.
-            model.zero_grad(set_to_none=(dp is None))
-            if dp is not None:
+            # set_to_none saves a fill + an accumulate add per param; under
+            # DP the bucketer zeroes its flat buffers instead (one fill per
+            # bucket, grads are views into them).
+            if dp is None:
+                model.zero_grad(set_to_none=True)
+            else:
                 dp.zero_grad()
             states = model.detach(states)
             with trace_range("forward"):
