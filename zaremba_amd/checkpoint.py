@@ -38,9 +38,10 @@ def save_checkpoint(path: str, model, epoch: int = 0, lr: float = 0.0,
         "epoch": epoch,
         "lr": lr,
         "torch_rng_state": torch.get_rng_state(),
-        # All visible device generators (the HIP dropout kernel draws its
-        # philox seed from python's secrets, re-seeded per process, but
-        # eager-path dropout and any torch.cuda sampling use these).
+        # All visible device generators. (The HIP dropout seed derives
+        # from the HOST torch generator at HipModel construction, so the
+        # torch_rng_state above governs it; these cover eager-path
+        # dropout and any torch.cuda sampling.)
         "cuda_rng_state": (torch.cuda.get_rng_state_all()
                            if torch.cuda.is_available() else None),
     }
