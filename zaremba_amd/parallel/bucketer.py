@@ -56,10 +56,17 @@ class _Bucket:
 
 
 class GradBucketer:
-    def __init__(self, model: torch.nn.Module, bucket_bytes: int = 25 << 20,
+    def __init__(self, model: torch.nn.Module, bucket_bytes: int = None,
                  comm_dtype=None):
         if not (td.is_available() and td.is_initialized()):
             raise RuntimeError("GradBucketer requires an initialized process group")
+        if bucket_bytes is None:
+            import os
+            # ZAREMBA_AMD_BUCKET_MB tunes overlap granularity on real
+            # multi-GPU nodes (default 25 MB: a handful of in-flight
+            # collectives without launch-bound slivers)
+            bucket_bytes = int(float(
+                os.environ.get("ZAREMBA_AMD_BUCKET_MB", "25")) * (1 << 20))
         if comm_dtype is None:
             import os
             comm_dtype = (torch.float32
