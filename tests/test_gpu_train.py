@@ -207,3 +207,17 @@ def test_seeded_dropout_reproducible():
     a = run()
     b = run()
     assert torch.equal(a, b)
+
+
+def test_hip_engine_rejects_large_batch():
+    """batch_size > 32 is outside the 32-row MFMA tiling contract; the
+    HIP engine must raise instead of corrupting memory."""
+    import pytest as _pytest
+    import torch
+    from zaremba_amd.models.lstm_lm import Model
+
+    m = Model(50, 64, 1, dropout=0.0, winit=0.1, engine="hip").to(dev())
+    x = torch.randint(0, 50, (4, 40), device=dev())
+    s = m.state_init(40)
+    with _pytest.raises(RuntimeError, match="batch_size <= 32"):
+        m(x, s)

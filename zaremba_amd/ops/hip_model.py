@@ -285,6 +285,15 @@ class HipModel:
             self.refresh_shadows()
         m = self.model
         T, B = x.shape
+        if B > 32:
+            # Every cell kernel (persistent, fused backward, per-step
+            # fallback) tiles the batch into the 32-row MFMA pair;
+            # larger per-GPU batches are out of contract — fail loudly
+            # instead of corrupting memory.
+            raise RuntimeError(
+                f"HIP engine supports batch_size <= 32 per GPU (32-row "
+                f"MFMA tiling); got {B}. Use --engine eager, or shard "
+                f"the batch across GPUs with data parallelism.")
         p = m.dropout_p
         idx = x.reshape(-1)
         H = m.hidden_size
