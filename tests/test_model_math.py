@@ -142,3 +142,28 @@ def test_dropout_fresh_mask_per_call():
     assert not torch.equal(a, b)
     model.eval()
     assert torch.equal(model.dropout(x), x)
+
+
+def test_lstm_layer_gradcheck_fp64():
+    """fp64 torch.autograd.gradcheck of the eager oracle cell (SURVEY §4:
+    the oracle every HIP kernel is validated against must itself have
+    exact gradients)."""
+    import torch
+    from zaremba_amd.ops import functional as F_ref
+
+    torch.manual_seed(0)
+    T, B, H = 3, 2, 4
+    x = torch.randn(T, B, H, dtype=torch.float64, requires_grad=True)
+    h0 = torch.randn(B, H, dtype=torch.float64, requires_grad=True)
+    c0 = torch.randn(B, H, dtype=torch.float64, requires_grad=True)
+    Wx = torch.randn(4 * H, H, dtype=torch.float64, requires_grad=True)
+    Wh = torch.randn(4 * H, H, dtype=torch.float64, requires_grad=True)
+    bx = torch.randn(4 * H, dtype=torch.float64, requires_grad=True)
+    bh = torch.randn(4 * H, dtype=torch.float64, requires_grad=True)
+
+    def f(x, h0, c0, Wx, Wh, bx, bh):
+        out, h, c = F_ref.lstm_layer(x, h0, c0, Wx, Wh, bx, bh)
+        return out.sum() + h.sum() + c.sum()
+
+    assert torch.autograd.gradcheck(f, (x, h0, c0, Wx, Wh, bx, bh),
+                                    eps=1e-6, atol=1e-8)
