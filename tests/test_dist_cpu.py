@@ -14,7 +14,6 @@ import torch
 import torch.distributed as td
 import torch.multiprocessing as mp
 
-from zaremba_amd import trainer
 from zaremba_amd.models.lstm_lm import Model
 from zaremba_amd.ops import functional as F_ref
 

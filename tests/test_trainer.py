@@ -1,8 +1,6 @@
 """Trainer-driver semantics: LR schedule off-by-one, training convergence
 on a tiny synthetic task, perplexity evaluation, checkpoint round-trip."""
 
-import os
-
 import numpy as np
 import pytest
 import torch

@@ -26,7 +26,6 @@ import numpy as np
 import torch
 
 from . import _C
-from .ops import functional as F_ref
 
 
 def _prob_sum(models: Dict[str, torch.nn.Module], states, x, acc):
