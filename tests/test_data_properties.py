@@ -45,6 +45,13 @@ def test_minibatch_properties(n, B, T):
     if rows - 1 < T + 1:
         assert batches == [] or batches[-1][0].size(0) < T
 
+    # EXACT window-count oracle, transcribed from the reference loop
+    # (main.py:68-73): window at i iff min(T, rows-1-i) < rows-1-i,
+    # i.e. i + T < rows - 1 (deep-fuzzed at 5000 examples in-round)
+    expected = sum(1 for i in range(0, max(rows - 1, 0), T)
+                   if i + T < rows - 1)
+    assert len(batches) == expected, (n, B, T, len(batches), expected)
+
 
 @settings(max_examples=100, deadline=None)
 @given(
